@@ -1,0 +1,193 @@
+#!/usr/bin/env python3
+"""msbn flagship benchmark: ResNet-50 + SyncBatchNorm, bf16, synthetic 224x224
+ImageNet-shaped data (BASELINE.json config 3 — the headline metric:
+images/sec whole-node at 1/2/4/8 MI355X).
+
+Single GPU:      python bench.py --gpus 1 --steps 30 --warmup 10
+Multi GPU (driver): python -m torch.distributed.run --nnodes=1
+                    --nproc-per-node N --master-addr 127.0.0.1 bench.py
+                    --gpus N --steps K --warmup W
+
+One JSON line is printed by rank 0 (contract in the project brief).
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+
+import msbn
+
+
+def cast_bf16_keep_bn_fp32(model):
+    """bf16 weights everywhere except BatchNorm affine params + running stats
+    (standard mixed-precision practice; msbn kernels take bf16 activations
+    with fp32 channel parameters)."""
+    from msbn.nn.batchnorm import _NormBase
+
+    model.to(torch.bfloat16)
+    for m in model.modules():
+        if isinstance(m, (_NormBase, torch.nn.modules.batchnorm._BatchNorm)):
+            m.float()
+    return model
+
+
+def build_model(name, memory_format, dtype, device, sync):
+    if name == "resnet50":
+        model = msbn.models.resnet50()
+    elif name == "resnet18":
+        model = msbn.models.resnet18()
+    else:
+        raise ValueError(name)
+    if sync:
+        model = msbn.convert_sync_batchnorm(model)
+    model = model.to(device)
+    if dtype == torch.bfloat16:
+        model = cast_bf16_keep_bn_fp32(model)
+    if memory_format == torch.channels_last:
+        model = model.to(memory_format=torch.channels_last)
+    return model
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=30)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--batch-size", type=int, default=256,
+                   help="per-GPU batch (weak scaling)")
+    p.add_argument("--model", type=str, default="resnet50")
+    p.add_argument("--memory-format", type=str, default="channels_last",
+                   choices=["channels_last", "contiguous"])
+    p.add_argument("--dtype", type=str, default="bf16",
+                   choices=["bf16", "fp32"])
+    p.add_argument("--stock", action="store_true",
+                   help="use stock torch SyncBatchNorm+DDP (comparison line)")
+    p.add_argument("--local_rank", "--local-rank", type=int,
+                   default=int(os.environ.get("LOCAL_RANK", 0)),
+                   dest="local_rank")
+    args = p.parse_args()
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    distributed = world_size > 1
+    use_cuda = torch.cuda.is_available()
+    device = torch.device(f"cuda:{args.local_rank}" if use_cuda else "cpu")
+    if use_cuda:
+        torch.cuda.set_device(device)
+    if distributed:
+        dist.init_process_group(
+            backend="nccl" if use_cuda else "gloo", init_method="env://",
+            world_size=world_size, rank=rank,
+        )
+
+    dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
+    mem_fmt = (torch.channels_last if args.memory_format == "channels_last"
+               else torch.contiguous_format)
+
+    torch.manual_seed(1234)
+    if args.stock:
+        model = msbn.models.resnet50() if args.model == "resnet50" else \
+            msbn.models.resnet18()
+        if distributed:
+            model = torch.nn.SyncBatchNorm.convert_sync_batchnorm(model)
+        model = model.to(device)
+        if dtype == torch.bfloat16:
+            model = cast_bf16_keep_bn_fp32(model)
+        if mem_fmt == torch.channels_last:
+            model = model.to(memory_format=torch.channels_last)
+        if distributed:
+            model = torch.nn.parallel.DistributedDataParallel(
+                model, device_ids=[args.local_rank] if use_cuda else None,
+                output_device=args.local_rank if use_cuda else None,
+            )
+    else:
+        model = build_model(args.model, mem_fmt, dtype, device, sync=True)
+        if distributed:
+            model = msbn.parallel.DistributedDataParallel(
+                model,
+                device_ids=[args.local_rank] if use_cuda else None,
+                output_device=args.local_rank if use_cuda else None,
+                gradient_as_bucket_view=True,
+            )
+    model.train()
+
+    opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9,
+                          weight_decay=1e-4)
+    loss_fn = torch.nn.CrossEntropyLoss()
+
+    bs = args.batch_size
+    x = torch.randn(bs, 3, 224, 224, device=device, dtype=dtype)
+    if mem_fmt == torch.channels_last:
+        x = x.to(memory_format=torch.channels_last)
+    y = torch.randint(0, 1000, (bs,), device=device)
+
+    def step():
+        opt.zero_grad(set_to_none=True)
+        out = model(x)
+        loss = loss_fn(out.float(), y)
+        loss.backward()
+        opt.step()
+        return loss
+
+    for _ in range(args.warmup):
+        step()
+
+    if distributed:
+        dist.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    if use_cuda:
+        torch.cuda.synchronize()
+    if distributed:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # MAX elapsed over ranks decides whole-job throughput
+    if distributed:
+        t = torch.tensor([elapsed], device=device if use_cuda else "cpu",
+                         dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = t.item()
+
+    n_gpus = world_size
+    images_per_sec = bs * n_gpus * args.steps / elapsed
+    ms_per_step = 1000.0 * elapsed / args.steps
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": f"images/sec (whole node) {args.model} SyncBN",
+            "value": round(images_per_sec, 2),
+            "unit": "images/sec",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": args.dtype,
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": bs * n_gpus,
+                "per_gpu_batch": bs,
+                "image": "3x224x224",
+                "memory_format": args.memory_format,
+                "parallelism": f"dp{n_gpus}",
+                "impl": "stock" if args.stock else "msbn",
+            },
+        }))
+
+    if distributed:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

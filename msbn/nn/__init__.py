@@ -1,0 +1,17 @@
+from msbn.nn.batchnorm import (  # noqa: F401
+    BatchNorm1d,
+    BatchNorm2d,
+    BatchNorm3d,
+    SyncBatchNorm,
+    convert_sync_batchnorm,
+)
+from msbn.nn.functions import SyncBatchNormFunction  # noqa: F401
+
+__all__ = [
+    "BatchNorm1d",
+    "BatchNorm2d",
+    "BatchNorm3d",
+    "SyncBatchNorm",
+    "convert_sync_batchnorm",
+    "SyncBatchNormFunction",
+]

@@ -1,0 +1,163 @@
+"""The distributed SyncBatchNorm autograd function.
+
+Algorithm (behavioral parity with the stock sync path, SURVEY.md §2.2
+"_functions.py:7-209", re-designed MI355X-first):
+
+forward:
+  1. local per-channel moments via ONE fused kernel writing the packed
+     [mean | invstd | count] (2C+1 fp32) buffer directly (no cat()).
+  2. all_gather of the packed buffer over RCCL/xGMI -> [W, 2C+1].
+  3. fused gather kernel: zero-count masking IN-KERNEL (no GPU->CPU sync,
+     graph-capture safe), Chan variance merge with heterogeneous counts,
+     running-stats update (unbiased, momentum) in the same launch.
+  4. fused normalize-scale-shift (elemt kernel).
+
+backward:
+  1. fused per-channel reduction -> packed [sum_dy | sum_dy_xmu] (2C fp32)
+     plus grad_weight/grad_bias.
+  2. ONE all_reduce(SUM) of the packed 2C buffer.
+  3. fused grad_input kernel with the global element count.
+  grad_weight / grad_bias are NOT synced here — the DDP reducer's gradient
+  buckets carry them (same division of labor as stock, _functions.py:180-181).
+
+Empty-input ranks still post both collectives so peers are never blocked.
+"""
+
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+from msbn import ops
+
+
+def _is_nccl_like(process_group) -> bool:
+    try:
+        return dist.get_backend(process_group) in ("nccl", "hccl")
+    except Exception:
+        return False
+
+
+class SyncBatchNormFunction(torch.autograd.Function):
+    @staticmethod
+    def forward(
+        ctx,
+        input: torch.Tensor,
+        weight: Optional[torch.Tensor],
+        bias: Optional[torch.Tensor],
+        running_mean: Optional[torch.Tensor],
+        running_var: Optional[torch.Tensor],
+        eps: float,
+        momentum: float,
+        process_group,
+        world_size: int,
+    ):
+        if not (
+            input.is_contiguous(memory_format=torch.channels_last)
+            or input.is_contiguous(memory_format=torch.channels_last_3d)
+        ):
+            input = input.contiguous()
+        if weight is not None:
+            weight = weight.contiguous()
+
+        C = int(input.shape[1])
+        local_count = input.numel() // C if C > 0 else 0
+
+        if world_size > 1:
+            # -- packed local stats ---------------------------------------
+            packed = torch.empty(2 * C + 1, dtype=torch.float32, device=input.device)
+            if local_count > 0:
+                ops.batch_norm_stats_packed(input, eps, packed)
+            else:
+                packed.zero_()
+
+            # -- cross-replica gather over RCCL/xGMI (S5) ------------------
+            packed_all = torch.empty(
+                (world_size, 2 * C + 1), dtype=torch.float32, device=input.device
+            )
+            if _is_nccl_like(process_group):
+                dist.all_gather_into_tensor(packed_all, packed, group=process_group)
+            else:
+                chunks = list(packed_all.unbind(0))
+                dist.all_gather(chunks, packed, group=process_group)
+                packed_all = torch.stack(chunks, dim=0)
+
+            # -- combine + running-stats update (in-kernel masking) --------
+            mean, invstd, count_sum = ops.batch_norm_gather_stats_packed(
+                input, packed_all, running_mean, running_var, momentum, eps
+            )
+        else:
+            mean, invstd = ops.batch_norm_stats(input, eps)
+            count_sum = torch.full(
+                (1,), float(local_count), dtype=torch.float32, device=input.device
+            )
+            if running_mean is not None and local_count > 0:
+                with torch.no_grad():
+                    var = invstd.to(torch.float32).pow(-2) - eps
+                    unbiased = (
+                        var * (local_count / (local_count - 1.0))
+                        if local_count > 1
+                        else var
+                    )
+                    running_mean.mul_(1 - momentum).add_(
+                        mean.to(running_mean.dtype), alpha=momentum
+                    )
+                    running_var.mul_(1 - momentum).add_(
+                        unbiased.to(running_var.dtype), alpha=momentum
+                    )
+
+        ctx.save_for_backward(input, weight, mean, invstd, count_sum)
+        ctx.process_group = process_group
+        ctx.world_size = world_size
+
+        if local_count == 0:
+            return torch.empty_like(input)
+        return ops.batch_norm_elemt(input, weight, bias, mean, invstd, eps)
+
+    @staticmethod
+    def backward(ctx, grad_output: torch.Tensor):
+        if not (
+            grad_output.is_contiguous(memory_format=torch.channels_last)
+            or grad_output.is_contiguous(memory_format=torch.channels_last_3d)
+        ):
+            grad_output = grad_output.contiguous()
+        input, weight, mean, invstd, count_sum = ctx.saved_tensors
+        process_group = ctx.process_group
+        world_size = ctx.world_size
+        need_input_g, need_weight_g, need_bias_g = ctx.needs_input_grad[0:3]
+
+        C = int(input.shape[1])
+        local_count = input.numel() // C if C > 0 else 0
+        grad_input = grad_weight = grad_bias = None
+
+        if local_count > 0:
+            sum_dy, sum_dy_xmu, grad_weight, grad_bias = ops.batch_norm_backward_reduce(
+                grad_output, input, mean, invstd, weight,
+                need_input_g, need_weight_g, need_bias_g,
+            )
+            if need_input_g:
+                if world_size > 1:
+                    combined = torch.cat([sum_dy, sum_dy_xmu])
+                    dist.all_reduce(
+                        combined, dist.ReduceOp.SUM, group=process_group
+                    )
+                    sum_dy, sum_dy_xmu = combined[:C], combined[C:]
+                grad_input = ops.batch_norm_backward_elemt(
+                    grad_output, input, mean, invstd, weight,
+                    sum_dy, sum_dy_xmu, count_sum,
+                )
+        elif world_size > 1 and need_input_g:
+            # Empty-input rank: still contribute zeros to unblock peers
+            # (stock behavior, _functions.py:187-204).
+            combined = torch.zeros(
+                2 * C, dtype=torch.float32, device=grad_output.device
+            )
+            dist.all_reduce(combined, dist.ReduceOp.SUM, group=process_group)
+            grad_input = torch.empty_like(grad_output)
+
+        return (
+            grad_input,
+            grad_weight if need_weight_g else None,
+            grad_bias if need_bias_g else None,
+            None, None, None, None, None, None,
+        )

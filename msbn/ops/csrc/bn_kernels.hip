@@ -1,0 +1,1039 @@
+// msbn BatchNorm kernels for MI355X (gfx950, CDNA4).
+//
+// Replaces the stock SyncBatchNorm kernel family K1-K10 with
+// an MI355X-first design:
+//
+//  * Reductions are TWO-STAGE: a `partial` kernel with a grid sized for the
+//    256-CU chip (the stock one-block-per-channel shape leaves 3/4 of the chip
+//    idle at C=64) writes per-chunk fp64 {sum, sumsq} pairs to a workspace; a
+//    tiny `finalize` kernel combines them.  Deterministic (no atomics), and
+//    fp64 accumulation kills the E[x^2]-E[x]^2 cancellation risk outright.
+//  * wave64 shuffle reductions + LDS partials inside each block.
+//  * 16 B/lane vectorized loads (Pack<T,V>) for bf16/fp16/fp32, NCHW and
+//    channels-last (NHWC) layouts both first-class.
+//  * The sync path's cross-rank message is built IN the finalize kernel
+//    ([mean | invstd | count] packed fp32), and the combine kernel masks
+//    zero-count ranks in-device: no cat(), no GPU->CPU sync, hipGraph-safe.
+//  * Elementwise kernels apply per-channel affine coefficients precomputed by
+//    tiny kernels: y = x*scale+shift, dx = a*dy + b*x + d  (grid-stride,
+//    capped grid, fused multiply-add form).
+#include "bn_ops.hpp"
+#include "common.hpp"
+
+#include <ATen/ATen.h>
+#include <c10/hip/HIPStream.h>
+
+#include <algorithm>
+
+namespace msbn {
+
+namespace {
+
+// =====================================================================
+// stats partial: NCHW   grid(C, nchunkN, nchunkS) block 256
+// ws layout: [nchunk][C][2] fp64, chunk = by*gridDim.z + bz
+// =====================================================================
+template <typename T, int V>
+__global__ void bn_stats_partial_nchw(const T* __restrict__ x,
+                                      double* __restrict__ ws, int64_t N,
+                                      int64_t C, int64_t S, int64_t chunkN,
+                                      int64_t chunkS) {
+  const int64_t c = blockIdx.x;
+  const int64_t n0 = blockIdx.y * chunkN;
+  const int64_t n1 = i64min(n0 + chunkN, N);
+  const int64_t s0 = blockIdx.z * chunkS;
+  const int64_t s1 = i64min(s0 + chunkS, S);
+  double a = 0.0, b = 0.0;
+  for (int64_t n = n0; n < n1; ++n) {
+    const T* row = x + (n * C + c) * S;
+    if (V == 1) {
+      for (int64_t s = s0 + threadIdx.x; s < s1; s += blockDim.x) {
+        float v = to_f(row[s]);
+        a += v;
+        b += (double)v * v;
+      }
+    } else {
+      for (int64_t s = s0 + (int64_t)threadIdx.x * V; s < s1;
+           s += (int64_t)blockDim.x * V) {
+        Pack<T, V> pk = *reinterpret_cast<const Pack<T, V>*>(&row[s]);
+#pragma unroll
+        for (int k = 0; k < V; ++k) {
+          float v = to_f(pk.v[k]);
+          a += v;
+          b += (double)v * v;
+        }
+      }
+    }
+  }
+  __shared__ double lds[2 * (MSBN_BLOCK / MSBN_WAVE)];
+  block_reduce_pair(a, b, lds);
+  if (threadIdx.x == 0) {
+    const int64_t chunk = (int64_t)blockIdx.y * gridDim.z + blockIdx.z;
+    double* out = ws + (chunk * C + c) * 2;
+    out[0] = a;
+    out[1] = b;
+  }
+}
+
+// =====================================================================
+// stats partial: NHWC   grid(ctiles, nchunk) block 256
+// Each block covers channels [tile*lpr*V, ...) and a chunk of rows; thread
+// (lane, rowoff) owns V consecutive channels.  LDS tree-reduce over rowoff.
+// =====================================================================
+template <typename T, int V>
+__global__ void bn_stats_partial_nhwc(const T* __restrict__ x,
+                                      double* __restrict__ ws, int64_t rows,
+                                      int64_t C, int64_t chunk_rows, int lpr) {
+  const int rpi = blockDim.x / lpr;  // rows in flight per iteration
+  const int lane = threadIdx.x % lpr;
+  const int rowoff = threadIdx.x / lpr;
+  const bool active = rowoff < rpi;
+  const int64_t c = (int64_t)blockIdx.x * lpr * V + (int64_t)lane * V;
+  const bool inb = active && (c + V <= C);
+
+  double a[V], b[V];
+#pragma unroll
+  for (int k = 0; k < V; ++k) a[k] = b[k] = 0.0;
+
+  if (inb) {
+    const int64_t r0 = (int64_t)blockIdx.y * chunk_rows;
+    const int64_t r1 = i64min(r0 + chunk_rows, rows);
+    for (int64_t r = r0 + rowoff; r < r1; r += rpi) {
+      Pack<T, V> pk = *reinterpret_cast<const Pack<T, V>*>(&x[r * C + c]);
+#pragma unroll
+      for (int k = 0; k < V; ++k) {
+        float v = to_f(pk.v[k]);
+        a[k] += v;
+        b[k] += (double)v * v;
+      }
+    }
+  }
+
+  // LDS tree-reduce across the rowoff dimension (generic, non-pow2 safe).
+  __shared__ double sdata[MSBN_BLOCK * V * 2];
+#pragma unroll
+  for (int k = 0; k < V; ++k) {
+    sdata[(threadIdx.x * V + k) * 2] = a[k];
+    sdata[(threadIdx.x * V + k) * 2 + 1] = b[k];
+  }
+  for (int st = 1; st < rpi; st <<= 1) {
+    __syncthreads();
+    if (active && (rowoff & ((st << 1) - 1)) == 0 && rowoff + st < rpi) {
+      const int other = threadIdx.x + st * lpr;
+#pragma unroll
+      for (int k = 0; k < V; ++k) {
+        sdata[(threadIdx.x * V + k) * 2] += sdata[(other * V + k) * 2];
+        sdata[(threadIdx.x * V + k) * 2 + 1] += sdata[(other * V + k) * 2 + 1];
+      }
+    }
+  }
+  __syncthreads();
+  if (rowoff == 0 && c < C) {
+    const int64_t chunk = blockIdx.y;
+#pragma unroll
+    for (int k = 0; k < V; ++k) {
+      if (c + k < C) {
+        double* out = ws + (chunk * C + c + k) * 2;
+        out[0] = sdata[(threadIdx.x * V + k) * 2];
+        out[1] = sdata[(threadIdx.x * V + k) * 2 + 1];
+      }
+    }
+  }
+}
+
+// =====================================================================
+// stats finalize: combine chunk partials -> mean/invstd (+count, +running)
+// =====================================================================
+template <typename RT>
+__global__ void bn_stats_finalize(const double* __restrict__ ws, int nchunks,
+                                  int64_t C, double count, float eps,
+                                  float* __restrict__ mean,
+                                  float* __restrict__ invstd,
+                                  float* __restrict__ count_out,
+                                  RT* __restrict__ rmean, RT* __restrict__ rvar,
+                                  float momentum) {
+  const int64_t c = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  double a = 0.0, b = 0.0;
+  for (int ch = 0; ch < nchunks; ++ch) {
+    a += ws[((int64_t)ch * C + c) * 2];
+    b += ws[((int64_t)ch * C + c) * 2 + 1];
+  }
+  const double m = a / count;
+  double var = b / count - m * m;
+  var = var > 0.0 ? var : 0.0;
+  mean[c] = (float)m;
+  invstd[c] = (float)rsqrt(var + (double)eps);
+  if (c == 0 && count_out != nullptr) count_out[0] = (float)count;
+  if (rmean != nullptr) {
+    const double unbiased = count > 1.0 ? var * (count / (count - 1.0)) : var;
+    rmean[c] = from_f<RT>((1.f - momentum) * to_f(rmean[c]) + momentum * (float)m);
+    rvar[c] =
+        from_f<RT>((1.f - momentum) * to_f(rvar[c]) + momentum * (float)unbiased);
+  }
+}
+
+// =====================================================================
+// gather: combine W ranks' packed [mean | invstd | count] rows.
+// Zero-count ranks masked HERE (device-side; no host sync).
+// =====================================================================
+template <typename RT>
+__global__ void bn_gather_stats(const float* __restrict__ packed_all, int W,
+                                int64_t C, float eps, float momentum,
+                                float* __restrict__ mean,
+                                float* __restrict__ invstd,
+                                float* __restrict__ count_out,
+                                RT* __restrict__ rmean, RT* __restrict__ rvar) {
+  const int64_t c = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const int64_t row = 2 * C + 1;
+  double n_tot = 0.0, m_acc = 0.0, ex2 = 0.0;
+  for (int w = 0; w < W; ++w) {
+    const float cnt = packed_all[w * row + 2 * C];
+    if (cnt > 0.f) {
+      const double m = packed_all[w * row + c];
+      const double istd = packed_all[w * row + C + c];
+      const double var = 1.0 / (istd * istd) - (double)eps;
+      n_tot += cnt;
+      m_acc += cnt * m;
+      ex2 += cnt * (var + m * m);
+    }
+  }
+  if (n_tot == 0.0) {
+    mean[c] = 0.f;
+    invstd[c] = 0.f;
+    if (c == 0 && count_out != nullptr) count_out[0] = 0.f;
+    return;
+  }
+  const double m_g = m_acc / n_tot;
+  double var_g = ex2 / n_tot - m_g * m_g;
+  var_g = var_g > 0.0 ? var_g : 0.0;
+  mean[c] = (float)m_g;
+  invstd[c] = (float)rsqrt(var_g + (double)eps);
+  if (c == 0 && count_out != nullptr) count_out[0] = (float)n_tot;
+  if (rmean != nullptr) {
+    const double unbiased =
+        n_tot > 1.0 ? var_g * (n_tot / (n_tot - 1.0)) : var_g;
+    rmean[c] =
+        from_f<RT>((1.f - momentum) * to_f(rmean[c]) + momentum * (float)m_g);
+    rvar[c] = from_f<RT>((1.f - momentum) * to_f(rvar[c]) +
+                         momentum * (float)unbiased);
+  }
+}
+
+// =====================================================================
+// per-channel affine coefficient kernels (tiny)
+// =====================================================================
+template <typename WT>
+__global__ void bn_affine_fwd(const float* __restrict__ mean,
+                              const float* __restrict__ invstd,
+                              const WT* __restrict__ w,
+                              const WT* __restrict__ b, int64_t C,
+                              float* __restrict__ scale,
+                              float* __restrict__ shift) {
+  const int64_t c = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const float sc = invstd[c] * (w != nullptr ? to_f(w[c]) : 1.f);
+  scale[c] = sc;
+  shift[c] = -mean[c] * sc + (b != nullptr ? to_f(b[c]) : 0.f);
+}
+
+// dx = a*dy + b*x + d with
+//   f1 = invstd*gamma, f2 = sum_dy/n, f3 = invstd^2*sum_dy_xmu/n
+//   a = f1, b = -f1*f3, d = f1*(f3*mean - f2)
+template <typename WT>
+__global__ void bn_affine_bwd(const float* __restrict__ mean,
+                              const float* __restrict__ invstd,
+                              const WT* __restrict__ w,
+                              const float* __restrict__ sum_dy,
+                              const float* __restrict__ sum_dy_xmu,
+                              const float* __restrict__ count, int64_t C,
+                              float* __restrict__ coef_a,
+                              float* __restrict__ coef_b,
+                              float* __restrict__ coef_d) {
+  const int64_t c = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const float n = count[0];
+  if (n <= 0.f) {
+    coef_a[c] = coef_b[c] = coef_d[c] = 0.f;
+    return;
+  }
+  const float istd = invstd[c];
+  const float f1 = istd * (w != nullptr ? to_f(w[c]) : 1.f);
+  const float f2 = sum_dy[c] / n;
+  const float f3 = istd * istd * sum_dy_xmu[c] / n;
+  coef_a[c] = f1;
+  coef_b[c] = -f1 * f3;
+  coef_d[c] = f1 * (f3 * mean[c] - f2);
+}
+
+// =====================================================================
+// elementwise: y = x*scale[c] + shift[c]
+// =====================================================================
+template <typename T, int V>
+__global__ void bn_elemt_nchw(const T* __restrict__ x, T* __restrict__ y,
+                              const float* __restrict__ scale,
+                              const float* __restrict__ shift, int64_t total,
+                              int64_t C, int64_t S) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i * V < total; i += stride) {
+    const int64_t e = i * V;
+    const int64_t c = (e / S) % C;
+    const float sc = scale[c], sh = shift[c];
+    if (V == 1) {
+      y[e] = from_f<T>(to_f(x[e]) * sc + sh);
+    } else {
+      Pack<T, V> px = *reinterpret_cast<const Pack<T, V>*>(&x[e]);
+      Pack<T, V> py;
+#pragma unroll
+      for (int k = 0; k < V; ++k) py.v[k] = from_f<T>(to_f(px.v[k]) * sc + sh);
+      *reinterpret_cast<Pack<T, V>*>(&y[e]) = py;
+    }
+  }
+}
+
+template <typename T, int V>
+__global__ void bn_elemt_nhwc(const T* __restrict__ x, T* __restrict__ y,
+                              const float* __restrict__ scale,
+                              const float* __restrict__ shift, int64_t total,
+                              int64_t C) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i * V < total; i += stride) {
+    const int64_t e = i * V;
+    const int64_t c = e % C;
+    if (V == 1) {
+      y[e] = from_f<T>(to_f(x[e]) * scale[c] + shift[c]);
+    } else {
+      Pack<T, V> px = *reinterpret_cast<const Pack<T, V>*>(&x[e]);
+      Pack<float, V> ps = *reinterpret_cast<const Pack<float, V>*>(&scale[c]);
+      Pack<float, V> pb = *reinterpret_cast<const Pack<float, V>*>(&shift[c]);
+      Pack<T, V> py;
+#pragma unroll
+      for (int k = 0; k < V; ++k)
+        py.v[k] = from_f<T>(to_f(px.v[k]) * ps.v[k] + pb.v[k]);
+      *reinterpret_cast<Pack<T, V>*>(&y[e]) = py;
+    }
+  }
+}
+
+// =====================================================================
+// backward reduce partial: {sum_dy, sum_dy*(x-mean)} per channel
+// =====================================================================
+template <typename T, int V>
+__global__ void bn_bwd_reduce_partial_nchw(const T* __restrict__ dy,
+                                           const T* __restrict__ x,
+                                           const float* __restrict__ mean,
+                                           double* __restrict__ ws, int64_t N,
+                                           int64_t C, int64_t S, int64_t chunkN,
+                                           int64_t chunkS) {
+  const int64_t c = blockIdx.x;
+  const float m = mean[c];
+  const int64_t n0 = blockIdx.y * chunkN;
+  const int64_t n1 = i64min(n0 + chunkN, N);
+  const int64_t s0 = blockIdx.z * chunkS;
+  const int64_t s1 = i64min(s0 + chunkS, S);
+  double a = 0.0, b = 0.0;
+  for (int64_t n = n0; n < n1; ++n) {
+    const int64_t base = (n * C + c) * S;
+    if (V == 1) {
+      for (int64_t s = s0 + threadIdx.x; s < s1; s += blockDim.x) {
+        const float g = to_f(dy[base + s]);
+        a += g;
+        b += (double)g * (to_f(x[base + s]) - m);
+      }
+    } else {
+      for (int64_t s = s0 + (int64_t)threadIdx.x * V; s < s1;
+           s += (int64_t)blockDim.x * V) {
+        Pack<T, V> pg = *reinterpret_cast<const Pack<T, V>*>(&dy[base + s]);
+        Pack<T, V> px = *reinterpret_cast<const Pack<T, V>*>(&x[base + s]);
+#pragma unroll
+        for (int k = 0; k < V; ++k) {
+          const float g = to_f(pg.v[k]);
+          a += g;
+          b += (double)g * (to_f(px.v[k]) - m);
+        }
+      }
+    }
+  }
+  __shared__ double lds[2 * (MSBN_BLOCK / MSBN_WAVE)];
+  block_reduce_pair(a, b, lds);
+  if (threadIdx.x == 0) {
+    const int64_t chunk = (int64_t)blockIdx.y * gridDim.z + blockIdx.z;
+    double* out = ws + (chunk * C + c) * 2;
+    out[0] = a;
+    out[1] = b;
+  }
+}
+
+template <typename T, int V>
+__global__ void bn_bwd_reduce_partial_nhwc(const T* __restrict__ dy,
+                                           const T* __restrict__ x,
+                                           const float* __restrict__ mean,
+                                           double* __restrict__ ws,
+                                           int64_t rows, int64_t C,
+                                           int64_t chunk_rows, int lpr) {
+  const int rpi = blockDim.x / lpr;
+  const int lane = threadIdx.x % lpr;
+  const int rowoff = threadIdx.x / lpr;
+  const bool active = rowoff < rpi;
+  const int64_t c = (int64_t)blockIdx.x * lpr * V + (int64_t)lane * V;
+  const bool inb = active && (c + V <= C);
+
+  double a[V], b[V];
+  float m[V];
+#pragma unroll
+  for (int k = 0; k < V; ++k) {
+    a[k] = b[k] = 0.0;
+    m[k] = 0.f;
+  }
+  if (inb) {
+#pragma unroll
+    for (int k = 0; k < V; ++k) m[k] = mean[c + k];
+    const int64_t r0 = (int64_t)blockIdx.y * chunk_rows;
+    const int64_t r1 = i64min(r0 + chunk_rows, rows);
+    for (int64_t r = r0 + rowoff; r < r1; r += rpi) {
+      Pack<T, V> pg = *reinterpret_cast<const Pack<T, V>*>(&dy[r * C + c]);
+      Pack<T, V> px = *reinterpret_cast<const Pack<T, V>*>(&x[r * C + c]);
+#pragma unroll
+      for (int k = 0; k < V; ++k) {
+        const float g = to_f(pg.v[k]);
+        a[k] += g;
+        b[k] += (double)g * (to_f(px.v[k]) - m[k]);
+      }
+    }
+  }
+  __shared__ double sdata[MSBN_BLOCK * V * 2];
+#pragma unroll
+  for (int k = 0; k < V; ++k) {
+    sdata[(threadIdx.x * V + k) * 2] = a[k];
+    sdata[(threadIdx.x * V + k) * 2 + 1] = b[k];
+  }
+  for (int st = 1; st < rpi; st <<= 1) {
+    __syncthreads();
+    if (active && (rowoff & ((st << 1) - 1)) == 0 && rowoff + st < rpi) {
+      const int other = threadIdx.x + st * lpr;
+#pragma unroll
+      for (int k = 0; k < V; ++k) {
+        sdata[(threadIdx.x * V + k) * 2] += sdata[(other * V + k) * 2];
+        sdata[(threadIdx.x * V + k) * 2 + 1] += sdata[(other * V + k) * 2 + 1];
+      }
+    }
+  }
+  __syncthreads();
+  if (rowoff == 0 && c < C) {
+    const int64_t chunk = blockIdx.y;
+#pragma unroll
+    for (int k = 0; k < V; ++k) {
+      if (c + k < C) {
+        double* out = ws + (chunk * C + c + k) * 2;
+        out[0] = sdata[(threadIdx.x * V + k) * 2];
+        out[1] = sdata[(threadIdx.x * V + k) * 2 + 1];
+      }
+    }
+  }
+}
+
+template <typename WT>
+__global__ void bn_bwd_reduce_finalize(const double* __restrict__ ws,
+                                       int nchunks, int64_t C,
+                                       const float* __restrict__ invstd,
+                                       float* __restrict__ sum_dy,
+                                       float* __restrict__ sum_dy_xmu,
+                                       WT* __restrict__ grad_weight,
+                                       WT* __restrict__ grad_bias) {
+  const int64_t c = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  double a = 0.0, b = 0.0;
+  for (int ch = 0; ch < nchunks; ++ch) {
+    a += ws[((int64_t)ch * C + c) * 2];
+    b += ws[((int64_t)ch * C + c) * 2 + 1];
+  }
+  if (sum_dy != nullptr) sum_dy[c] = (float)a;
+  if (sum_dy_xmu != nullptr) sum_dy_xmu[c] = (float)b;
+  if (grad_weight != nullptr) grad_weight[c] = from_f<WT>((float)(b * invstd[c]));
+  if (grad_bias != nullptr) grad_bias[c] = from_f<WT>((float)a);
+}
+
+// =====================================================================
+// backward elementwise: dx = a[c]*dy + b[c]*x + d[c]
+// =====================================================================
+template <typename T, int V>
+__global__ void bn_bwd_elemt_nchw(const T* __restrict__ dy,
+                                  const T* __restrict__ x, T* __restrict__ dx,
+                                  const float* __restrict__ ca,
+                                  const float* __restrict__ cb,
+                                  const float* __restrict__ cd, int64_t total,
+                                  int64_t C, int64_t S) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i * V < total; i += stride) {
+    const int64_t e = i * V;
+    const int64_t c = (e / S) % C;
+    const float A = ca[c], B = cb[c], D = cd[c];
+    if (V == 1) {
+      dx[e] = from_f<T>(A * to_f(dy[e]) + B * to_f(x[e]) + D);
+    } else {
+      Pack<T, V> pg = *reinterpret_cast<const Pack<T, V>*>(&dy[e]);
+      Pack<T, V> px = *reinterpret_cast<const Pack<T, V>*>(&x[e]);
+      Pack<T, V> po;
+#pragma unroll
+      for (int k = 0; k < V; ++k)
+        po.v[k] = from_f<T>(A * to_f(pg.v[k]) + B * to_f(px.v[k]) + D);
+      *reinterpret_cast<Pack<T, V>*>(&dx[e]) = po;
+    }
+  }
+}
+
+template <typename T, int V>
+__global__ void bn_bwd_elemt_nhwc(const T* __restrict__ dy,
+                                  const T* __restrict__ x, T* __restrict__ dx,
+                                  const float* __restrict__ ca,
+                                  const float* __restrict__ cb,
+                                  const float* __restrict__ cd, int64_t total,
+                                  int64_t C) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i * V < total; i += stride) {
+    const int64_t e = i * V;
+    const int64_t c = e % C;
+    if (V == 1) {
+      dx[e] = from_f<T>(ca[c] * to_f(dy[e]) + cb[c] * to_f(x[e]) + cd[c]);
+    } else {
+      Pack<T, V> pg = *reinterpret_cast<const Pack<T, V>*>(&dy[e]);
+      Pack<T, V> px = *reinterpret_cast<const Pack<T, V>*>(&x[e]);
+      Pack<float, V> pa = *reinterpret_cast<const Pack<float, V>*>(&ca[c]);
+      Pack<float, V> pb = *reinterpret_cast<const Pack<float, V>*>(&cb[c]);
+      Pack<float, V> pd = *reinterpret_cast<const Pack<float, V>*>(&cd[c]);
+      Pack<T, V> po;
+#pragma unroll
+      for (int k = 0; k < V; ++k)
+        po.v[k] =
+            from_f<T>(pa.v[k] * to_f(pg.v[k]) + pb.v[k] * to_f(px.v[k]) + pd.v[k]);
+      *reinterpret_cast<Pack<T, V>*>(&dx[e]) = po;
+    }
+  }
+}
+
+// =====================================================================
+// host-side helpers
+// =====================================================================
+
+struct Layout {
+  bool nhwc;       // channels-last (rows x C contiguous)
+  int64_t N, C, S; // NCHW view (S = spatial)
+  int64_t rows;    // NHWC view (= N*S)
+};
+
+Layout get_layout(const at::Tensor& t) {
+  Layout L{};
+  L.C = t.size(1);
+  L.N = t.size(0);
+  L.S = t.numel() / std::max<int64_t>(L.N * L.C, 1);
+  L.rows = L.N * L.S;
+  const auto fmt = t.suggest_memory_format();
+  if ((fmt == at::MemoryFormat::ChannelsLast && t.dim() == 4 &&
+       t.is_contiguous(at::MemoryFormat::ChannelsLast)) ||
+      (fmt == at::MemoryFormat::ChannelsLast3d && t.dim() == 5 &&
+       t.is_contiguous(at::MemoryFormat::ChannelsLast3d))) {
+    L.nhwc = true;
+  } else {
+    TORCH_CHECK(t.is_contiguous(), "msbn: input must be contiguous (NCHW) or "
+                                   "channels-last");
+    L.nhwc = false;
+  }
+  return L;
+}
+
+inline int64_t cdiv(int64_t a, int64_t b) { return (a + b - 1) / b; }
+
+// pick vector width: 16B/lane when layout & alignment allow
+template <typename T>
+int pick_v(const void* p0, const void* p1, const void* p2, int64_t inner) {
+  const int vmax = 16 / (int)sizeof(T);
+  auto ok = [&](int v) {
+    const size_t bytes = (size_t)v * sizeof(T);
+    auto aligned = [&](const void* p) {
+      return p == nullptr || ((uintptr_t)p % bytes) == 0;
+    };
+    return inner % v == 0 && aligned(p0) && aligned(p1) && aligned(p2);
+  };
+  if (ok(vmax)) return vmax;
+  if (vmax >= 4 && ok(vmax / 2)) return vmax / 2;
+  return 1;
+}
+
+constexpr int kTargetBlocks = 2048;  // 256 CUs x 8 blocks (guide G11)
+
+struct NchwGrid {
+  dim3 grid;
+  int64_t chunkN, chunkS;
+  int nchunks;
+};
+
+NchwGrid nchw_grid(int64_t N, int64_t C, int64_t S, int V) {
+  NchwGrid g{};
+  const int64_t perC = std::max<int64_t>(1, kTargetBlocks / std::max<int64_t>(C, 1));
+  int64_t nchunkN = std::min<int64_t>(N, perC);
+  nchunkN = std::max<int64_t>(nchunkN, 1);
+  g.chunkN = cdiv(N, nchunkN);
+  nchunkN = cdiv(N, g.chunkN);
+  int64_t want_s = cdiv(perC, nchunkN);
+  // each S-chunk should be a multiple of V and >= ~4096 elements of work
+  int64_t max_s = cdiv(S, std::max<int64_t>((int64_t)MSBN_BLOCK * V, 1024));
+  int64_t nchunkS = std::max<int64_t>(1, std::min(want_s, std::max<int64_t>(max_s, 1)));
+  g.chunkS = cdiv(cdiv(S, nchunkS), V) * V;
+  nchunkS = cdiv(S, g.chunkS);
+  g.grid = dim3((unsigned)C, (unsigned)nchunkN, (unsigned)nchunkS);
+  g.nchunks = (int)(nchunkN * nchunkS);
+  return g;
+}
+
+struct NhwcGrid {
+  dim3 grid;
+  int64_t chunk_rows;
+  int nchunks;
+  int lpr;
+};
+
+NhwcGrid nhwc_grid(int64_t rows, int64_t C, int V) {
+  NhwcGrid g{};
+  const int64_t cv = cdiv(C, V);
+  g.lpr = (int)std::min<int64_t>(cv, MSBN_BLOCK);
+  const int64_t ctiles = cdiv(C, (int64_t)g.lpr * V);
+  const int64_t per_tile =
+      std::max<int64_t>(1, kTargetBlocks / std::max<int64_t>(ctiles, 1));
+  const int rpi = MSBN_BLOCK / g.lpr;
+  int64_t max_chunks = cdiv(rows, std::max<int64_t>(4 * rpi, 16));
+  int64_t nchunks = std::max<int64_t>(1, std::min(per_tile, std::max<int64_t>(max_chunks, 1)));
+  g.chunk_rows = cdiv(rows, nchunks);
+  nchunks = cdiv(rows, g.chunk_rows);
+  g.grid = dim3((unsigned)ctiles, (unsigned)nchunks);
+  g.nchunks = (int)nchunks;
+  return g;
+}
+
+int elemt_grid(int64_t total, int V) {
+  return (int)std::min<int64_t>(cdiv(total, (int64_t)MSBN_BLOCK * V),
+                                2 * kTargetBlocks);
+}
+
+#define MSBN_DISPATCH_FLOAT_TYPES(TYPE, NAME, ...)                          \
+  [&] {                                                                     \
+    switch (TYPE) {                                                         \
+      case at::kFloat: {                                                    \
+        using scalar_t = float;                                             \
+        using native_t = float;                                             \
+        return __VA_ARGS__();                                               \
+      }                                                                     \
+      case at::kBFloat16: {                                                 \
+        using scalar_t = at::BFloat16;                                      \
+        using native_t = __hip_bfloat16;                                    \
+        return __VA_ARGS__();                                               \
+      }                                                                     \
+      case at::kHalf: {                                                     \
+        using scalar_t = at::Half;                                          \
+        using native_t = __half;                                            \
+        return __VA_ARGS__();                                               \
+      }                                                                     \
+      default:                                                              \
+        TORCH_CHECK(false, NAME, ": unsupported dtype ", TYPE);             \
+    }                                                                       \
+  }()
+
+#define MSBN_DISPATCH_V(VSEL, VMAX, ...)                                    \
+  [&] {                                                                     \
+    if constexpr (VMAX == 8) {                                              \
+      switch (VSEL) {                                                       \
+        case 8: {                                                           \
+          constexpr int VV = 8;                                             \
+          return __VA_ARGS__();                                             \
+        }                                                                   \
+        case 4: {                                                           \
+          constexpr int VV = 4;                                             \
+          return __VA_ARGS__();                                             \
+        }                                                                   \
+        default: {                                                          \
+          constexpr int VV = 1;                                             \
+          return __VA_ARGS__();                                             \
+        }                                                                   \
+      }                                                                     \
+    } else {                                                                \
+      switch (VSEL) {                                                       \
+        case 4: {                                                           \
+          constexpr int VV = 4;                                             \
+          return __VA_ARGS__();                                             \
+        }                                                                   \
+        case 2: {                                                           \
+          constexpr int VV = 2;                                             \
+          return __VA_ARGS__();                                             \
+        }                                                                   \
+        default: {                                                          \
+          constexpr int VV = 1;                                             \
+          return __VA_ARGS__();                                             \
+        }                                                                   \
+      }                                                                     \
+    }                                                                       \
+  }()
+
+#define MSBN_DISPATCH_RSTAT(TYPE, NAME, ...)                                \
+  [&] {                                                                     \
+    switch (TYPE) {                                                         \
+      case at::kFloat: {                                                    \
+        using rstat_t = float;                                              \
+        return __VA_ARGS__();                                               \
+      }                                                                     \
+      case at::kBFloat16: {                                                 \
+        using rstat_t = __hip_bfloat16;                                     \
+        return __VA_ARGS__();                                               \
+      }                                                                     \
+      case at::kHalf: {                                                     \
+        using rstat_t = __half;                                             \
+        return __VA_ARGS__();                                               \
+      }                                                                     \
+      default:                                                              \
+        TORCH_CHECK(false, NAME, ": unsupported running-stat dtype ", TYPE); \
+    }                                                                       \
+  }()
+
+hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+// Launch the two-stage stats reduction; writes mean/invstd/count into the
+// given fp32 pointers (which may alias a packed buffer).
+void stats_into(const at::Tensor& input, double eps, float* mean_p,
+                float* invstd_p, float* count_p, const at::Tensor* rmean,
+                const at::Tensor* rvar, double momentum) {
+  const Layout L = get_layout(input);
+  const int64_t count = L.rows;
+  auto stream = cur_stream();
+  const auto rstat_type = rmean ? rmean->scalar_type() : at::kFloat;
+
+  MSBN_DISPATCH_FLOAT_TYPES(input.scalar_type(), "batch_norm_stats", [&] {
+    const native_t* x =
+        reinterpret_cast<const native_t*>(input.data_ptr<scalar_t>());
+    constexpr int VMAX = 16 / (int)sizeof(native_t);
+    at::Tensor ws;
+    int nchunks = 0;
+    if (!L.nhwc) {
+      const int v = pick_v<native_t>(x, nullptr, nullptr, L.S);
+      auto g = nchw_grid(L.N, L.C, L.S, v);
+      nchunks = g.nchunks;
+      ws = at::empty({(int64_t)nchunks * L.C * 2},
+                     input.options().dtype(at::kDouble));
+      MSBN_DISPATCH_V(v, VMAX, [&] {
+        hipLaunchKernelGGL((bn_stats_partial_nchw<native_t, VV>), g.grid,
+                           dim3(MSBN_BLOCK), 0, stream, x,
+                           ws.data_ptr<double>(), L.N, L.C, L.S, g.chunkN,
+                           g.chunkS);
+      });
+    } else {
+      const int v = pick_v<native_t>(x, nullptr, nullptr, L.C);
+      auto g = nhwc_grid(L.rows, L.C, v);
+      nchunks = g.nchunks;
+      ws = at::empty({(int64_t)nchunks * L.C * 2},
+                     input.options().dtype(at::kDouble));
+      MSBN_DISPATCH_V(v, VMAX, [&] {
+        hipLaunchKernelGGL((bn_stats_partial_nhwc<native_t, VV>), g.grid,
+                           dim3(MSBN_BLOCK), 0, stream, x,
+                           ws.data_ptr<double>(), L.rows, L.C, g.chunk_rows,
+                           g.lpr);
+      });
+    }
+    const int fgrid = (int)cdiv(L.C, MSBN_BLOCK);
+    MSBN_DISPATCH_RSTAT(rstat_type, "batch_norm_stats", [&] {
+      rstat_t* rm = rmean ? reinterpret_cast<rstat_t*>(rmean->data_ptr())
+                          : nullptr;
+      rstat_t* rv = rvar ? reinterpret_cast<rstat_t*>(rvar->data_ptr())
+                         : nullptr;
+      hipLaunchKernelGGL((bn_stats_finalize<rstat_t>), dim3(fgrid),
+                         dim3(MSBN_BLOCK), 0, stream, ws.data_ptr<double>(),
+                         nchunks, L.C, (double)count, (float)eps, mean_p,
+                         invstd_p, count_p, rm, rv, (float)momentum);
+    });
+  });
+}
+
+}  // namespace
+
+// =====================================================================
+// public ops
+// =====================================================================
+
+std::tuple<at::Tensor, at::Tensor> batch_norm_stats(const at::Tensor& input,
+                                                    double eps) {
+  const int64_t C = input.size(1);
+  auto opts = input.options().dtype(at::kFloat);
+  auto mean = at::empty({C}, opts);
+  auto invstd = at::empty({C}, opts);
+  if (input.numel() == 0) {
+    mean.zero_();
+    invstd.zero_();
+    return {mean, invstd};
+  }
+  stats_into(input, eps, mean.data_ptr<float>(), invstd.data_ptr<float>(),
+             nullptr, nullptr, nullptr, 0.0);
+  return {mean, invstd};
+}
+
+void batch_norm_stats_packed(const at::Tensor& input, double eps,
+                             at::Tensor& out) {
+  const int64_t C = input.size(1);
+  TORCH_CHECK(out.is_contiguous() && out.numel() == 2 * C + 1 &&
+                  out.scalar_type() == at::kFloat,
+              "packed stats buffer must be contiguous fp32 of size 2C+1");
+  if (input.numel() == 0) {
+    out.zero_();
+    return;
+  }
+  float* p = out.data_ptr<float>();
+  stats_into(input, eps, p, p + C, p + 2 * C, nullptr, nullptr, 0.0);
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> batch_norm_gather_stats_packed(
+    const at::Tensor& packed_all, const c10::optional<at::Tensor>& running_mean,
+    const c10::optional<at::Tensor>& running_var, double momentum, double eps) {
+  TORCH_CHECK(packed_all.dim() == 2 && packed_all.is_contiguous() &&
+                  packed_all.scalar_type() == at::kFloat,
+              "packed_all must be contiguous fp32 [W, 2C+1]");
+  const int W = (int)packed_all.size(0);
+  const int64_t C = (packed_all.size(1) - 1) / 2;
+  auto opts = packed_all.options();
+  auto mean = at::empty({C}, opts);
+  auto invstd = at::empty({C}, opts);
+  auto count_sum = at::empty({1}, opts);
+  auto stream = cur_stream();
+  const auto rstat_type =
+      running_mean.has_value() ? running_mean->scalar_type() : at::kFloat;
+  const int fgrid = (int)cdiv(C, MSBN_BLOCK);
+  MSBN_DISPATCH_RSTAT(rstat_type, "gather_stats", [&] {
+    rstat_t* rm = running_mean.has_value()
+                      ? reinterpret_cast<rstat_t*>(running_mean->data_ptr())
+                      : nullptr;
+    rstat_t* rv = running_var.has_value()
+                      ? reinterpret_cast<rstat_t*>(running_var->data_ptr())
+                      : nullptr;
+    hipLaunchKernelGGL((bn_gather_stats<rstat_t>), dim3(fgrid),
+                       dim3(MSBN_BLOCK), 0, stream,
+                       packed_all.data_ptr<float>(), W, C, (float)eps,
+                       (float)momentum, mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(), count_sum.data_ptr<float>(),
+                       rm, rv);
+  });
+  return {mean, invstd, count_sum};
+}
+
+std::tuple<at::Tensor, at::Tensor> batch_norm_gather_stats_with_counts(
+    const at::Tensor& mean_all, const at::Tensor& invstd_all,
+    const c10::optional<at::Tensor>& running_mean,
+    const c10::optional<at::Tensor>& running_var, double momentum, double eps,
+    const at::Tensor& counts) {
+  // API-parity wrapper: assemble the packed layout, then run the packed path.
+  const int64_t W = mean_all.size(0);
+  const int64_t C = mean_all.size(1);
+  auto packed = at::empty({W, 2 * C + 1}, mean_all.options().dtype(at::kFloat));
+  packed.narrow(1, 0, C).copy_(mean_all);
+  packed.narrow(1, C, C).copy_(invstd_all);
+  packed.narrow(1, 2 * C, 1).copy_(
+      counts.to(at::kFloat).reshape({W, 1}));
+  auto out = batch_norm_gather_stats_packed(packed, running_mean, running_var,
+                                            momentum, eps);
+  return {std::get<0>(out), std::get<1>(out)};
+}
+
+at::Tensor batch_norm_elemt(const at::Tensor& input,
+                            const c10::optional<at::Tensor>& weight,
+                            const c10::optional<at::Tensor>& bias,
+                            const at::Tensor& mean, const at::Tensor& invstd,
+                            double eps) {
+  (void)eps;  // invstd already folds eps
+  const Layout L = get_layout(input);
+  auto out = at::empty_like(input);
+  if (input.numel() == 0) return out;
+  auto stream = cur_stream();
+  auto f32 = input.options().dtype(at::kFloat);
+  auto coefs = at::empty({2 * L.C}, f32);
+  float* scale = coefs.data_ptr<float>();
+  float* shift = scale + L.C;
+  const int cgrid = (int)cdiv(L.C, MSBN_BLOCK);
+  const auto wtype = weight.has_value() ? weight->scalar_type()
+                     : bias.has_value() ? bias->scalar_type()
+                                        : at::kFloat;
+  MSBN_DISPATCH_RSTAT(wtype, "bn_elemt", [&] {
+    const rstat_t* w =
+        weight.has_value()
+            ? reinterpret_cast<const rstat_t*>(weight->data_ptr())
+            : nullptr;
+    const rstat_t* b = bias.has_value()
+                           ? reinterpret_cast<const rstat_t*>(bias->data_ptr())
+                           : nullptr;
+    hipLaunchKernelGGL((bn_affine_fwd<rstat_t>), dim3(cgrid), dim3(MSBN_BLOCK),
+                       0, stream, mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(), w, b, L.C, scale, shift);
+  });
+
+  const int64_t total = input.numel();
+  MSBN_DISPATCH_FLOAT_TYPES(input.scalar_type(), "bn_elemt", [&] {
+    const native_t* x =
+        reinterpret_cast<const native_t*>(input.data_ptr<scalar_t>());
+    native_t* y = reinterpret_cast<native_t*>(out.data_ptr<scalar_t>());
+    constexpr int VMAX = 16 / (int)sizeof(native_t);
+    if (!L.nhwc) {
+      const int v = pick_v<native_t>(x, y, nullptr, L.S);
+      const int grid = elemt_grid(total, v);
+      MSBN_DISPATCH_V(v, VMAX, [&] {
+        hipLaunchKernelGGL((bn_elemt_nchw<native_t, VV>), dim3(grid),
+                           dim3(MSBN_BLOCK), 0, stream, x, y, scale, shift,
+                           total, L.C, L.S);
+      });
+    } else {
+      const int v = pick_v<native_t>(x, y, nullptr, L.C);
+      const int grid = elemt_grid(total, v);
+      MSBN_DISPATCH_V(v, VMAX, [&] {
+        hipLaunchKernelGGL((bn_elemt_nhwc<native_t, VV>), dim3(grid),
+                           dim3(MSBN_BLOCK), 0, stream, x, y, scale, shift,
+                           total, L.C);
+      });
+    }
+  });
+  return out;
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor>
+batch_norm_backward_reduce(const at::Tensor& grad_out, const at::Tensor& input,
+                           const at::Tensor& mean, const at::Tensor& invstd,
+                           const c10::optional<at::Tensor>& weight, bool input_g,
+                           bool weight_g, bool bias_g) {
+  const Layout L = get_layout(input);
+  TORCH_CHECK(grad_out.sizes() == input.sizes(), "grad_out/input shape mismatch");
+  auto f32 = input.options().dtype(at::kFloat);
+  // ONE contiguous buffer for [sum_dy | sum_dy_xmu] -> single all_reduce.
+  auto combined = at::empty({2 * L.C}, f32);
+  auto sum_dy = combined.narrow(0, 0, L.C);
+  auto sum_dy_xmu = combined.narrow(0, L.C, L.C);
+  const auto wtype =
+      weight.has_value() ? weight->scalar_type() : input.scalar_type();
+  auto wopts = input.options().dtype(wtype);
+  at::Tensor grad_weight, grad_bias;
+  if (weight_g) grad_weight = at::empty({L.C}, wopts);
+  if (bias_g) grad_bias = at::empty({L.C}, wopts);
+
+  auto stream = cur_stream();
+  MSBN_DISPATCH_FLOAT_TYPES(input.scalar_type(), "bn_bwd_reduce", [&] {
+    const native_t* x =
+        reinterpret_cast<const native_t*>(input.data_ptr<scalar_t>());
+    const native_t* dy =
+        reinterpret_cast<const native_t*>(grad_out.data_ptr<scalar_t>());
+    constexpr int VMAX = 16 / (int)sizeof(native_t);
+    at::Tensor ws;
+    int nchunks = 0;
+    if (!L.nhwc) {
+      const int v = pick_v<native_t>(x, dy, nullptr, L.S);
+      auto g = nchw_grid(L.N, L.C, L.S, v);
+      nchunks = g.nchunks;
+      ws = at::empty({(int64_t)nchunks * L.C * 2},
+                     input.options().dtype(at::kDouble));
+      MSBN_DISPATCH_V(v, VMAX, [&] {
+        hipLaunchKernelGGL((bn_bwd_reduce_partial_nchw<native_t, VV>), g.grid,
+                           dim3(MSBN_BLOCK), 0, stream, dy, x,
+                           mean.data_ptr<float>(), ws.data_ptr<double>(), L.N,
+                           L.C, L.S, g.chunkN, g.chunkS);
+      });
+    } else {
+      const int v = pick_v<native_t>(x, dy, nullptr, L.C);
+      auto g = nhwc_grid(L.rows, L.C, v);
+      nchunks = g.nchunks;
+      ws = at::empty({(int64_t)nchunks * L.C * 2},
+                     input.options().dtype(at::kDouble));
+      MSBN_DISPATCH_V(v, VMAX, [&] {
+        hipLaunchKernelGGL((bn_bwd_reduce_partial_nhwc<native_t, VV>), g.grid,
+                           dim3(MSBN_BLOCK), 0, stream, dy, x,
+                           mean.data_ptr<float>(), ws.data_ptr<double>(),
+                           L.rows, L.C, g.chunk_rows, g.lpr);
+      });
+    }
+    const int fgrid = (int)cdiv(L.C, MSBN_BLOCK);
+    MSBN_DISPATCH_RSTAT(wtype, "bn_bwd_reduce", [&] {
+      rstat_t* gw = weight_g ? reinterpret_cast<rstat_t*>(grad_weight.data_ptr())
+                             : nullptr;
+      rstat_t* gb =
+          bias_g ? reinterpret_cast<rstat_t*>(grad_bias.data_ptr()) : nullptr;
+      hipLaunchKernelGGL((bn_bwd_reduce_finalize<rstat_t>), dim3(fgrid),
+                         dim3(MSBN_BLOCK), 0, stream, ws.data_ptr<double>(),
+                         nchunks, L.C, invstd.data_ptr<float>(),
+                         sum_dy.data_ptr<float>(),
+                         sum_dy_xmu.data_ptr<float>(), gw, gb);
+    });
+  });
+  (void)input_g;
+  return {sum_dy, sum_dy_xmu, grad_weight, grad_bias};
+}
+
+at::Tensor batch_norm_backward_elemt(
+    const at::Tensor& grad_out, const at::Tensor& input, const at::Tensor& mean,
+    const at::Tensor& invstd, const c10::optional<at::Tensor>& weight,
+    const at::Tensor& sum_dy, const at::Tensor& sum_dy_xmu,
+    const at::Tensor& count) {
+  const Layout L = get_layout(input);
+  auto dx = at::empty_like(grad_out);
+  if (input.numel() == 0) return dx;
+  // total count as a device fp32 scalar (no host sync)
+  at::Tensor count_sum;
+  if (count.numel() == 1 && count.scalar_type() == at::kFloat) {
+    count_sum = count;
+  } else {
+    count_sum = count.to(at::kFloat).sum().reshape({1});
+  }
+  auto stream = cur_stream();
+  auto f32 = input.options().dtype(at::kFloat);
+  auto coefs = at::empty({3 * L.C}, f32);
+  float* ca = coefs.data_ptr<float>();
+  float* cb = ca + L.C;
+  float* cd = cb + L.C;
+  const int cgrid = (int)cdiv(L.C, MSBN_BLOCK);
+  const auto wtype =
+      weight.has_value() ? weight->scalar_type() : at::kFloat;
+  MSBN_DISPATCH_RSTAT(wtype, "bn_bwd_elemt", [&] {
+    const rstat_t* w =
+        weight.has_value()
+            ? reinterpret_cast<const rstat_t*>(weight->data_ptr())
+            : nullptr;
+    hipLaunchKernelGGL((bn_affine_bwd<rstat_t>), dim3(cgrid), dim3(MSBN_BLOCK),
+                       0, stream, mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(), w, sum_dy.data_ptr<float>(),
+                       sum_dy_xmu.data_ptr<float>(),
+                       count_sum.data_ptr<float>(), L.C, ca, cb, cd);
+  });
+
+  const int64_t total = input.numel();
+  MSBN_DISPATCH_FLOAT_TYPES(input.scalar_type(), "bn_bwd_elemt", [&] {
+    const native_t* x =
+        reinterpret_cast<const native_t*>(input.data_ptr<scalar_t>());
+    const native_t* dy =
+        reinterpret_cast<const native_t*>(grad_out.data_ptr<scalar_t>());
+    native_t* o = reinterpret_cast<native_t*>(dx.data_ptr<scalar_t>());
+    constexpr int VMAX = 16 / (int)sizeof(native_t);
+    if (!L.nhwc) {
+      const int v = pick_v<native_t>(x, dy, o, L.S);
+      const int grid = elemt_grid(total, v);
+      MSBN_DISPATCH_V(v, VMAX, [&] {
+        hipLaunchKernelGGL((bn_bwd_elemt_nchw<native_t, VV>), dim3(grid),
+                           dim3(MSBN_BLOCK), 0, stream, dy, x, o, ca, cb, cd,
+                           total, L.C, L.S);
+      });
+    } else {
+      const int v = pick_v<native_t>(x, dy, o, L.C);
+      const int grid = elemt_grid(total, v);
+      MSBN_DISPATCH_V(v, VMAX, [&] {
+        hipLaunchKernelGGL((bn_bwd_elemt_nhwc<native_t, VV>), dim3(grid),
+                           dim3(MSBN_BLOCK), 0, stream, dy, x, o, ca, cb, cd,
+                           total, L.C);
+      });
+    }
+  });
+  return dx;
+}
+
+}  // namespace msbn

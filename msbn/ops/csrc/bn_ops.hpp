@@ -1,0 +1,58 @@
+// Host-side API of the msbn gfx950 BatchNorm kernel set (bn_kernels.hip).
+// Semantics: SURVEY.md §2.3 (the five stock SyncBatchNorm ATen ops), re-designed
+// for MI355X (two-stage deterministic fp64 reductions sized for 256 CUs,
+// NCHW + channels-last layouts, packed stat buffers for single-collective sync).
+#pragma once
+
+#include <ATen/ATen.h>
+
+#include <tuple>
+
+namespace msbn {
+
+// (mean, invstd) fp32 [C]; invstd = 1/sqrt(biased_var + eps).
+std::tuple<at::Tensor, at::Tensor> batch_norm_stats(const at::Tensor& input,
+                                                    double eps);
+
+// Fused: write [mean(C) | invstd(C) | count(1)] into `out` (fp32, 2C+1).
+void batch_norm_stats_packed(const at::Tensor& input, double eps,
+                             at::Tensor& out);
+
+// Combine per-rank stats; counts-weighted Chan merge; in-place running update.
+std::tuple<at::Tensor, at::Tensor> batch_norm_gather_stats_with_counts(
+    const at::Tensor& mean_all, const at::Tensor& invstd_all,
+    const c10::optional<at::Tensor>& running_mean,
+    const c10::optional<at::Tensor>& running_var, double momentum, double eps,
+    const at::Tensor& counts);
+
+// Packed variant: packed_all is [W, 2C+1]; returns (mean, invstd, count_sum[1]).
+// Zero-count ranks masked in-kernel (no GPU->CPU sync; hipGraph-safe).
+std::tuple<at::Tensor, at::Tensor, at::Tensor> batch_norm_gather_stats_packed(
+    const at::Tensor& packed_all, const c10::optional<at::Tensor>& running_mean,
+    const c10::optional<at::Tensor>& running_var, double momentum, double eps);
+
+// y = (x - mean) * invstd * weight + bias (scale/shift precomputed per channel).
+at::Tensor batch_norm_elemt(const at::Tensor& input,
+                            const c10::optional<at::Tensor>& weight,
+                            const c10::optional<at::Tensor>& bias,
+                            const at::Tensor& mean, const at::Tensor& invstd,
+                            double eps);
+
+// (sum_dy, sum_dy_xmu, grad_weight, grad_bias); sum_dy / sum_dy_xmu are views
+// into ONE contiguous [2C] fp32 buffer so the cross-rank all_reduce is a single
+// message (S6) with no cat() kernel.
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor>
+batch_norm_backward_reduce(const at::Tensor& grad_out, const at::Tensor& input,
+                           const at::Tensor& mean, const at::Tensor& invstd,
+                           const c10::optional<at::Tensor>& weight, bool input_g,
+                           bool weight_g, bool bias_g);
+
+// grad_input; `count` is a device tensor: [1] fp32 total count, or [W]
+// per-rank counts (any int/float dtype) which are summed on device.
+at::Tensor batch_norm_backward_elemt(
+    const at::Tensor& grad_out, const at::Tensor& input, const at::Tensor& mean,
+    const at::Tensor& invstd, const c10::optional<at::Tensor>& weight,
+    const at::Tensor& sum_dy, const at::Tensor& sum_dy_xmu,
+    const at::Tensor& count);
+
+}  // namespace msbn

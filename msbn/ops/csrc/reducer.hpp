@@ -1,0 +1,471 @@
+// msbn C++ DDP reducer — bucketed, backward-overlapped gradient all-reduce.
+//
+// MI355X-native equivalent of the stock c10d::Reducer (SURVEY.md §2.2
+// "reducer.hpp:45-581"), built directly on the c10d ProcessGroup C++ API
+// (RCCL over xGMI on ROCm; gloo on CPU):
+//
+//  * autograd post-accumulate hooks registered in C++ on each parameter's
+//    grad accumulator — they fire on the autograd engine's worker threads
+//    (no GIL, no Python) during backward.
+//  * gradients are copied into per-bucket flat buffers; when a bucket's
+//    pending count hits zero, buckets launch IN ORDER onto the process
+//    group, whose RCCL backend runs them on its dedicated HIP stream,
+//    event-gated against the compute stream -> comm overlaps the rest of
+//    backward (S7 in SURVEY.md §2.5).
+//  * finalize (queued on the engine via queue_callback from the first hook)
+//    waits the works and materializes averaged grads.
+//  * first-iteration gradient-arrival order is recorded; rebuild_buckets()
+//    re-bins parameters by that order (rank 0's order broadcast to all ranks)
+//    so buckets fill in backward order — the overlap lever the stock reducer
+//    calls _rebuild_buckets.
+#pragma once
+
+#include <torch/extension.h>
+
+#include <torch/csrc/autograd/engine.h>
+#include <torch/csrc/autograd/function.h>
+#include <torch/csrc/autograd/variable.h>
+#include <torch/csrc/autograd/utils/lambda_post_hook.h>
+#include <torch/csrc/distributed/c10d/ProcessGroup.hpp>
+#include <torch/csrc/distributed/c10d/Types.hpp>
+#include <torch/csrc/distributed/c10d/Work.hpp>
+
+#include <algorithm>
+#include <map>
+#include <memory>
+#include <mutex>
+#include <sstream>
+#include <vector>
+
+namespace msbn {
+
+using torch::autograd::utils::LambdaPostHook;
+using torch::autograd::variable_list;
+
+// ---------------------------------------------------------------------------
+// bucket assignment: greedy size-binning per (device, dtype) key, preserving
+// the given order (callers pass params in reverse registration order to
+// approximate backward order — stock behavior, distributed.py:1224-1230).
+// ---------------------------------------------------------------------------
+inline std::vector<std::vector<int64_t>> compute_bucket_assignment_by_size(
+    const std::vector<at::Tensor>& tensors,
+    const std::vector<size_t>& bucket_size_limits) {
+  TORCH_CHECK(!bucket_size_limits.empty(), "need at least one bucket limit");
+  struct Key {
+    c10::DeviceType dev_type;
+    c10::DeviceIndex dev_index;
+    at::ScalarType dtype;
+    bool operator<(const Key& o) const {
+      return std::tie(dev_type, dev_index, dtype) <
+             std::tie(o.dev_type, o.dev_index, o.dtype);
+    }
+  };
+  struct Bin {
+    std::vector<int64_t> indices;
+    size_t bytes = 0;
+    size_t limit_idx = 0;
+  };
+  std::map<Key, Bin> open;
+  std::vector<std::pair<int64_t, std::vector<int64_t>>> done;  // (first idx, bin)
+
+  for (int64_t i = 0; i < (int64_t)tensors.size(); ++i) {
+    const auto& t = tensors[i];
+    Key key{t.device().type(), t.device().index(), t.scalar_type()};
+    auto& bin = open[key];
+    const size_t bytes = (size_t)t.numel() * t.element_size();
+    const size_t limit =
+        bucket_size_limits[std::min(bin.limit_idx, bucket_size_limits.size() - 1)];
+    if (!bin.indices.empty() && bin.bytes + bytes > limit) {
+      done.emplace_back(bin.indices.front(), std::move(bin.indices));
+      bin.indices.clear();
+      bin.bytes = 0;
+      bin.limit_idx++;
+    }
+    bin.indices.push_back(i);
+    bin.bytes += bytes;
+  }
+  for (auto& kv : open) {
+    if (!kv.second.indices.empty()) {
+      done.emplace_back(kv.second.indices.front(),
+                        std::move(kv.second.indices));
+    }
+  }
+  // stable order: by first parameter index within the (reversed) list
+  std::sort(done.begin(), done.end(),
+            [](const auto& a, const auto& b) { return a.first < b.first; });
+  std::vector<std::vector<int64_t>> out;
+  out.reserve(done.size());
+  for (auto& d : done) out.push_back(std::move(d.second));
+  return out;
+}
+
+// ---------------------------------------------------------------------------
+// coalesced broadcast: flatten tensors into <= buffer_bytes chunks, broadcast
+// each from src, unflatten.  Used for init param sync (250 MiB) and the
+// per-iteration buffer sync (S4/S10 in SURVEY.md §2.5).
+// ---------------------------------------------------------------------------
+inline void broadcast_coalesced(const c10::intrusive_ptr<c10d::ProcessGroup>& pg,
+                         std::vector<at::Tensor> tensors, size_t buffer_bytes,
+                         int64_t src_rank) {
+  size_t i = 0;
+  while (i < tensors.size()) {
+    // take a run of tensors with same device+dtype up to buffer_bytes
+    size_t j = i;
+    size_t bytes = 0;
+    const auto dev = tensors[i].device();
+    const auto dt = tensors[i].scalar_type();
+    while (j < tensors.size() && tensors[j].device() == dev &&
+           tensors[j].scalar_type() == dt) {
+      const size_t b = (size_t)tensors[j].numel() * tensors[j].element_size();
+      if (j > i && bytes + b > buffer_bytes) break;
+      bytes += b;
+      ++j;
+    }
+    std::vector<at::Tensor> group(tensors.begin() + i, tensors.begin() + j);
+    at::Tensor flat;
+    if (group.size() == 1 && group[0].is_contiguous()) {
+      flat = group[0].reshape({-1});
+    } else {
+      std::vector<at::Tensor> flats;
+      flats.reserve(group.size());
+      for (auto& t : group) flats.push_back(t.reshape({-1}));
+      flat = at::cat(flats);
+    }
+    {
+      std::vector<at::Tensor> v{flat};
+      c10d::BroadcastOptions opts;
+      opts.rootRank = src_rank;
+      pg->broadcast(v, opts)->wait();
+    }
+    if (!(group.size() == 1 && group[0].is_contiguous())) {
+      int64_t off = 0;
+      for (auto& t : group) {
+        t.copy_(flat.narrow(0, off, t.numel()).view_as(t));
+        off += t.numel();
+      }
+    }
+    i = j;
+  }
+}
+
+// Broadcast rank-0 parameter metadata (ndim + sizes per param) and compare;
+// raises on any mismatch (stock _verify_param_shape_across_processes, S3).
+inline void verify_params_across_processes(
+    const c10::intrusive_ptr<c10d::ProcessGroup>& pg,
+    const std::vector<at::Tensor>& params) {
+  std::vector<int64_t> meta;
+  meta.push_back((int64_t)params.size());
+  for (const auto& p : params) {
+    meta.push_back(p.dim());
+    for (auto s : p.sizes()) meta.push_back(s);
+  }
+  auto opts = at::TensorOptions()
+                  .dtype(at::kLong)
+                  .device(params.empty() ? at::Device(at::kCPU)
+                                         : params[0].device());
+  auto local = at::tensor(meta, at::TensorOptions().dtype(at::kLong))
+                   .to(opts.device());
+  auto sz = at::tensor({(int64_t)meta.size()},
+                       at::TensorOptions().dtype(at::kLong))
+                .to(opts.device());
+  {
+    std::vector<at::Tensor> v{sz};
+    c10d::BroadcastOptions bo;
+    bo.rootRank = 0;
+    pg->broadcast(v, bo)->wait();
+  }
+  const int64_t root_len = sz.cpu().item<int64_t>();
+  TORCH_CHECK((int64_t)meta.size() == root_len,
+              "msbn DDP: parameter metadata length differs from rank 0 (",
+              meta.size(), " vs ", root_len,
+              ") — models differ across processes");
+  auto root = local.clone();
+  {
+    std::vector<at::Tensor> v{root};
+    c10d::BroadcastOptions bo;
+    bo.rootRank = 0;
+    pg->broadcast(v, bo)->wait();
+  }
+  TORCH_CHECK(root.cpu().equal(local.cpu()),
+              "msbn DDP: parameter shapes differ from rank 0 — all processes "
+              "must hold identical models");
+}
+
+// ---------------------------------------------------------------------------
+// Reducer
+// ---------------------------------------------------------------------------
+class Reducer : public std::enable_shared_from_this<Reducer> {
+ public:
+  Reducer(std::vector<at::Tensor> params,
+          std::vector<std::vector<int64_t>> bucket_indices,
+          c10::intrusive_ptr<c10d::ProcessGroup> pg, bool gradient_as_bucket_view,
+          size_t first_bucket_bytes, size_t bucket_bytes)
+      : params_(std::move(params)),
+        pg_(std::move(pg)),
+        gradient_as_bucket_view_(gradient_as_bucket_view),
+        first_bucket_bytes_(first_bucket_bytes),
+        bucket_bytes_(bucket_bytes),
+        div_factor_((double)pg_->getSize()) {
+    initialize_buckets(std::move(bucket_indices));
+    attach_hooks();
+  }
+
+  ~Reducer() { *alive_ = false; }
+
+  void prepare_for_backward(const std::vector<int64_t>& unused_params) {
+    std::lock_guard<std::mutex> lock(mutex_);
+    expect_autograd_hooks_ = true;
+    finalize_queued_ = false;
+    next_bucket_ = 0;
+    arrival_order_.clear();
+    for (auto& b : buckets_) {
+      b.pending = b.param_indices.size();
+      b.ready = false;
+      b.launched = false;
+    }
+    std::fill(param_ready_.begin(), param_ready_.end(), false);
+    // Params known-unused this iteration contribute zeros (find_unused path).
+    for (auto i : unused_params) {
+      zero_view_locked(i);
+      mark_ready_locked(i, /*from_hook=*/false);
+    }
+  }
+
+  void set_grad_sync_enabled(bool enabled) {
+    std::lock_guard<std::mutex> lock(mutex_);
+    sync_enabled_ = enabled;
+  }
+
+  // Called from the autograd post hook of parameter i.
+  void autograd_hook(int64_t i) {
+    std::lock_guard<std::mutex> lock(mutex_);
+    if (!expect_autograd_hooks_ || !sync_enabled_) return;
+    if (!param_ready_[i] && arrival_order_.size() < params_.size()) {
+      arrival_order_.push_back(i);
+    }
+    if (!finalize_queued_) {
+      finalize_queued_ = true;
+      auto self = shared_from_this();
+      torch::autograd::Engine::get_default_engine().queue_callback(
+          [self] { self->finalize_backward(); });
+    }
+    copy_grad_to_view_locked(i);
+    mark_ready_locked(i, /*from_hook=*/true);
+  }
+
+  void finalize_backward() {
+    std::lock_guard<std::mutex> lock(mutex_);
+    expect_autograd_hooks_ = false;
+    // every bucket must have launched
+    if (next_bucket_ != buckets_.size()) {
+      std::ostringstream oss;
+      oss << "msbn DDP: expected all gradient buckets to be ready at the end "
+             "of backward ("
+          << next_bucket_ << "/" << buckets_.size()
+          << " launched). Some parameters did not receive gradients; unused "
+             "parameter indices: [";
+      for (size_t i = 0; i < params_.size(); ++i)
+        if (!param_ready_[i]) oss << i << ", ";
+      oss << "]. Run the DDP wrapper with find_unused_parameters=True.";
+      finalize_queued_ = false;
+      TORCH_CHECK(false, oss.str());
+    }
+    for (auto& b : buckets_) {
+      if (b.work) {
+        b.work->wait();
+        b.work.reset();
+      }
+      for (size_t k = 0; k < b.param_indices.size(); ++k) {
+        auto& p = params_[b.param_indices[k]];
+        auto view = b.views[k];
+        auto grad = p.mutable_grad();
+        if (gradient_as_bucket_view_) {
+          if (!grad.defined() || grad.data_ptr() != view.data_ptr()) {
+            p.mutable_grad() = view.view_as(p);
+          }
+        } else {
+          if (grad.defined()) {
+            grad.copy_(view.view_as(p), /*non_blocking=*/true);
+          } else {
+            p.mutable_grad() = view.view_as(p).clone();
+          }
+        }
+      }
+    }
+    iterations_++;
+    finalize_queued_ = false;
+  }
+
+  // Re-bin parameters by the recorded first-iteration arrival order; rank 0's
+  // binning is broadcast so every rank allreduces identical buckets.
+  // Returns true if buckets changed.  Call OUTSIDE backward.
+  bool rebuild_buckets() {
+    std::unique_lock<std::mutex> lock(mutex_);
+    if (rebuilt_ || arrival_order_.size() != params_.size()) return false;
+    std::vector<int64_t> order = arrival_order_;
+    lock.unlock();
+
+    // broadcast rank 0's order
+    auto dev = params_[0].device();
+    auto t = at::tensor(order, at::TensorOptions().dtype(at::kLong)).to(dev);
+    {
+      std::vector<at::Tensor> v{t};
+      c10d::BroadcastOptions bo;
+      bo.rootRank = 0;
+      pg_->broadcast(v, bo)->wait();
+    }
+    auto tc = t.cpu();
+    auto acc = tc.accessor<int64_t, 1>();
+    std::vector<at::Tensor> ordered;
+    std::vector<int64_t> order2((size_t)acc.size(0));
+    ordered.reserve(params_.size());
+    for (int64_t k = 0; k < acc.size(0); ++k) {
+      order2[k] = acc[k];
+      ordered.push_back(params_[acc[k]]);
+    }
+    auto bins = compute_bucket_assignment_by_size(
+        ordered, {first_bucket_bytes_, bucket_bytes_});
+    // translate back to original param indices
+    for (auto& bin : bins)
+      for (auto& idx : bin) idx = order2[idx];
+
+    lock.lock();
+    initialize_buckets(std::move(bins));
+    rebuilt_ = true;
+    return true;
+  }
+
+  std::vector<std::vector<int64_t>> get_bucket_indices() const {
+    std::vector<std::vector<int64_t>> out;
+    for (auto& b : buckets_) out.push_back(b.param_indices);
+    return out;
+  }
+
+  int64_t iterations() const { return iterations_; }
+  bool rebuilt() const { return rebuilt_; }
+
+ private:
+  struct Bucket {
+    at::Tensor flat;
+    std::vector<at::Tensor> views;
+    std::vector<int64_t> param_indices;
+    size_t pending = 0;
+    bool ready = false;
+    bool launched = false;
+    c10::intrusive_ptr<c10d::Work> work;
+  };
+
+  void initialize_buckets(std::vector<std::vector<int64_t>> bucket_indices) {
+    buckets_.clear();
+    param_to_bucket_.assign(params_.size(), {-1, -1});
+    for (auto& idxs : bucket_indices) {
+      Bucket b;
+      b.param_indices = idxs;
+      int64_t total = 0;
+      for (auto i : idxs) total += params_[i].numel();
+      b.flat = at::zeros({total}, params_[idxs[0]].options());
+      int64_t off = 0;
+      for (size_t k = 0; k < idxs.size(); ++k) {
+        const auto n = params_[idxs[k]].numel();
+        b.views.push_back(b.flat.narrow(0, off, n));
+        param_to_bucket_[idxs[k]] = {(int64_t)buckets_.size(), (int64_t)k};
+        off += n;
+      }
+      b.pending = idxs.size();
+      buckets_.push_back(std::move(b));
+    }
+    param_ready_.assign(params_.size(), false);
+    for (size_t i = 0; i < params_.size(); ++i) {
+      TORCH_CHECK(param_to_bucket_[i].first >= 0,
+                  "msbn Reducer: parameter ", i, " missing from buckets");
+    }
+  }
+
+  void attach_hooks() {
+    grad_accumulators_.reserve(params_.size());
+    for (size_t i = 0; i < params_.size(); ++i) {
+      auto acc = torch::autograd::impl::grad_accumulator(params_[i]);
+      TORCH_CHECK(acc, "msbn Reducer: parameter ", i,
+                  " has no grad accumulator (requires_grad=False?)");
+      auto alive = alive_;
+      // `this` is safe while *alive: Reducer outlives hooks via DDP holder.
+      Reducer* self = this;
+      const int64_t idx = (int64_t)i;
+      acc->add_post_hook(std::make_unique<LambdaPostHook>(
+          [self, alive, idx](const variable_list& outputs,
+                             const variable_list& /*inputs*/) -> variable_list {
+            if (*alive) self->autograd_hook(idx);
+            return outputs;
+          }));
+      grad_accumulators_.push_back(std::move(acc));
+    }
+  }
+
+  void copy_grad_to_view_locked(int64_t i) {
+    auto [bi, vi] = param_to_bucket_[i];
+    auto& b = buckets_[bi];
+    auto view = b.views[vi];
+    auto& p = params_[i];
+    auto grad = p.grad();
+    if (!grad.defined()) {
+      view.zero_();
+      return;
+    }
+    if (grad.data_ptr() == view.data_ptr()) return;  // bucket-view grad
+    view.copy_(grad.reshape({-1}), /*non_blocking=*/true);
+  }
+
+  void zero_view_locked(int64_t i) {
+    auto [bi, vi] = param_to_bucket_[i];
+    buckets_[bi].views[vi].zero_();
+  }
+
+  void mark_ready_locked(int64_t i, bool from_hook) {
+    (void)from_hook;
+    if (param_ready_[i]) return;  // e.g. shared params firing twice
+    param_ready_[i] = true;
+    auto [bi, vi] = param_to_bucket_[i];
+    auto& b = buckets_[bi];
+    TORCH_CHECK(b.pending > 0, "msbn Reducer: bucket ", bi,
+                " marked ready too many times");
+    if (--b.pending == 0) b.ready = true;
+    // launch in fixed bucket order for cross-rank consistency
+    while (next_bucket_ < buckets_.size() && buckets_[next_bucket_].ready &&
+           !buckets_[next_bucket_].launched) {
+      launch_bucket_locked(buckets_[next_bucket_]);
+      next_bucket_++;
+    }
+  }
+
+  void launch_bucket_locked(Bucket& b) {
+    b.launched = true;
+    if (div_factor_ != 1.0) b.flat.div_(div_factor_);
+    std::vector<at::Tensor> v{b.flat};
+    c10d::AllreduceOptions opts;
+    b.work = pg_->allreduce(v, opts);
+  }
+
+  std::vector<at::Tensor> params_;
+  c10::intrusive_ptr<c10d::ProcessGroup> pg_;
+  bool gradient_as_bucket_view_;
+  size_t first_bucket_bytes_;
+  size_t bucket_bytes_;
+  double div_factor_;
+
+  std::vector<Bucket> buckets_;
+  std::vector<std::pair<int64_t, int64_t>> param_to_bucket_;
+  std::vector<bool> param_ready_;
+  std::vector<std::shared_ptr<torch::autograd::Node>> grad_accumulators_;
+  std::shared_ptr<bool> alive_ = std::make_shared<bool>(true);
+
+  std::mutex mutex_;
+  bool expect_autograd_hooks_ = false;
+  bool sync_enabled_ = true;
+  bool finalize_queued_ = false;
+  bool rebuilt_ = false;
+  size_t next_bucket_ = 0;
+  std::vector<int64_t> arrival_order_;
+  int64_t iterations_ = 0;
+};
+
+}  // namespace msbn

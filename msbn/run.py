@@ -1,0 +1,174 @@
+"""msbn.run — single/multi-node process launcher (torchrun-equivalent core).
+
+Spawns one worker process per GPU with the standard env contract
+(SURVEY.md §5.6 / run.py:187-232 of the stock launcher):
+
+  MASTER_ADDR, MASTER_PORT, RANK, LOCAL_RANK, WORLD_SIZE, LOCAL_WORLD_SIZE,
+  GROUP_RANK, NODE_RANK, TORCHELASTIC_RESTART_COUNT, TORCH_NCCL_ASYNC_ERROR_HANDLING=1
+
+plus optional ``--local-rank=<r>`` argv injection (legacy ``msbn.launch``
+contract, README.md:15-19).  On any worker failure the whole group is torn
+down and restarted (up to ``--max-restarts``), mirroring the elastic agent's
+restart-on-failure loop (SURVEY.md §2.2 "run.py:261-277").
+
+Usage:
+    python -m msbn.run --nproc-per-node=8 train.py ARGS...
+"""
+
+import argparse
+import os
+import signal
+import socket
+import subprocess
+import sys
+import time
+import uuid
+from typing import List, Optional
+
+
+def _free_port() -> int:
+    with socket.socket(socket.AF_INET, socket.SOCK_STREAM) as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+def _device_count() -> int:
+    try:
+        import torch
+
+        n = torch.cuda.device_count()
+        return n if n > 0 else 1
+    except Exception:
+        return 1
+
+
+def parse_args(argv: Optional[List[str]] = None, use_env_default: bool = True):
+    p = argparse.ArgumentParser(
+        description="msbn distributed launcher (one process per GPU)"
+    )
+    p.add_argument(
+        "--nproc-per-node", "--nproc_per_node", type=str, default="1",
+        help="number of workers on this node ('auto' = GPU count)",
+    )
+    p.add_argument("--nnodes", type=int, default=1)
+    p.add_argument("--node-rank", "--node_rank", type=int, default=0)
+    p.add_argument(
+        "--master-addr", "--master_addr", type=str, default="127.0.0.1"
+    )
+    p.add_argument("--master-port", "--master_port", type=int, default=None)
+    p.add_argument("--max-restarts", "--max_restarts", type=int, default=0)
+    p.add_argument(
+        "--use-env", "--use_env", action="store_true", default=use_env_default,
+        help="pass LOCAL_RANK via env only (no --local-rank argv)",
+    )
+    p.add_argument(
+        "--no-python", "--no_python", action="store_true",
+        help="run training_script directly instead of `python training_script`",
+    )
+    p.add_argument("--module", "-m", action="store_true",
+                   help="run the script as a python module (python -m script)")
+    p.add_argument("--run-id", "--run_id", type=str, default=None)
+    p.add_argument("training_script", type=str)
+    p.add_argument("training_script_args", nargs=argparse.REMAINDER)
+    return p.parse_args(argv)
+
+
+def _worker_env(args, local_rank: int, nproc: int, port: int, restart: int):
+    env = dict(os.environ)
+    world_size = nproc * args.nnodes
+    rank = args.node_rank * nproc + local_rank
+    env.update(
+        MASTER_ADDR=args.master_addr,
+        MASTER_PORT=str(port),
+        RANK=str(rank),
+        LOCAL_RANK=str(local_rank),
+        WORLD_SIZE=str(world_size),
+        LOCAL_WORLD_SIZE=str(nproc),
+        GROUP_RANK=str(args.node_rank),
+        NODE_RANK=str(args.node_rank),
+        TORCHELASTIC_RESTART_COUNT=str(restart),
+        TORCHELASTIC_RUN_ID=args.run_id or uuid.uuid4().hex[:8],
+        OMP_NUM_THREADS=env.get("OMP_NUM_THREADS", "1"),
+    )
+    env.setdefault("TORCH_NCCL_ASYNC_ERROR_HANDLING", "1")
+    return env
+
+
+def _spawn_group(args, nproc: int, port: int, restart: int):
+    procs = []
+    for local_rank in range(nproc):
+        cmd: List[str] = []
+        if not args.no_python:
+            cmd = [sys.executable, "-u"]
+            if args.module:
+                cmd.append("-m")
+        cmd.append(args.training_script)
+        script_args = list(args.training_script_args)
+        if not args.use_env:
+            script_args = [f"--local-rank={local_rank}"] + script_args
+        cmd.extend(script_args)
+        env = _worker_env(args, local_rank, nproc, port, restart)
+        procs.append(subprocess.Popen(cmd, env=env))
+    return procs
+
+
+def _kill_group(procs):
+    for p in procs:
+        if p.poll() is None:
+            p.send_signal(signal.SIGTERM)
+    deadline = time.time() + 10
+    for p in procs:
+        try:
+            p.wait(timeout=max(0.1, deadline - time.time()))
+        except subprocess.TimeoutExpired:
+            p.kill()
+
+
+def run(args) -> int:
+    nproc = (
+        _device_count()
+        if str(args.nproc_per_node) in ("auto", "gpu")
+        else int(args.nproc_per_node)
+    )
+    if args.run_id is None:
+        args.run_id = uuid.uuid4().hex[:8]
+    restarts = 0
+    while True:
+        port = args.master_port or _free_port()
+        procs = _spawn_group(args, nproc, port, restarts)
+        failed_rc = None
+        live = list(procs)
+        while live and failed_rc is None:
+            time.sleep(0.2)
+            for p in list(live):
+                rc = p.poll()
+                if rc is None:
+                    continue
+                live.remove(p)
+                if rc != 0:
+                    failed_rc = rc
+                    break
+        if failed_rc is None:
+            return 0
+        _kill_group(procs)
+        if restarts >= args.max_restarts:
+            print(
+                f"[msbn.run] worker failed with exit code {failed_rc}; "
+                f"no restarts left ({restarts}/{args.max_restarts})",
+                file=sys.stderr,
+            )
+            return failed_rc
+        restarts += 1
+        print(
+            f"[msbn.run] worker failed (rc={failed_rc}); restarting group "
+            f"({restarts}/{args.max_restarts})",
+            file=sys.stderr,
+        )
+
+
+def main(argv: Optional[List[str]] = None) -> int:
+    return run(parse_args(argv))
+
+
+if __name__ == "__main__":
+    sys.exit(main())

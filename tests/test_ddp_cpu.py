@@ -1,0 +1,283 @@
+"""Multi-process (gloo, world_size=2) integration tests on CPU.
+
+Pattern follows the stock test strategy: golden-model
+equivalence — a single-process model trained on the GLOBAL batch must match
+DDP+SyncBN ranks trained on per-rank SLICES, parameters compared after several
+optimizer steps (_test_DistributedDataParallel_SyncBatchNorm ≈
+distributed_test.py:5575-5634).
+"""
+
+import os
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+WORLD = 2
+
+
+def _worker(rank, fn_name, tmpdir, q):
+    try:
+        dist.init_process_group(
+            "gloo",
+            init_method=f"file://{tmpdir}/pg",
+            rank=rank,
+            world_size=WORLD,
+        )
+        fn = globals()[fn_name]
+        fn(rank)
+        q.put((rank, None))
+    except Exception as e:  # pragma: no cover
+        import traceback
+
+        q.put((rank, traceback.format_exc()))
+    finally:
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+def _spawn(fn_name, tmp_path):
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    ps = [
+        ctx.Process(target=_worker, args=(r, fn_name, str(tmp_path), q))
+        for r in range(WORLD)
+    ]
+    for p in ps:
+        p.start()
+    errs = []
+    for _ in range(WORLD):
+        rank, err = q.get()
+        if err:
+            errs.append((rank, err))
+    for p in ps:
+        p.join(timeout=60)
+        if p.is_alive():
+            p.terminate()
+            errs.append((p.pid, "timeout"))
+    assert not errs, "\n".join(f"rank {r}:\n{e}" for r, e in errs)
+
+
+# --------------------------------------------------------------------------
+def _golden_body(rank):
+    import msbn
+
+    torch.manual_seed(42)  # SAME init everywhere
+    global_bs = 8
+    local_bs = global_bs // WORLD
+
+    ddp_net = msbn.convert_sync_batchnorm(msbn.models.SimpleCNN(width=8))
+    ddp_net = msbn.parallel.DistributedDataParallel(ddp_net)
+    opt = torch.optim.SGD(ddp_net.parameters(), lr=0.05, momentum=0.9)
+
+    torch.manual_seed(42)
+    gold = msbn.models.SimpleCNN(width=8)  # plain BN on the global batch
+    gold_opt = torch.optim.SGD(gold.parameters(), lr=0.05, momentum=0.9)
+
+    loss_fn = torch.nn.CrossEntropyLoss()
+    for it in range(5):
+        g = torch.Generator().manual_seed(100 + it)
+        x = torch.randn(global_bs, 3, 16, 16, generator=g)
+        y = torch.randint(0, 10, (global_bs,), generator=g)
+        xs = x[rank * local_bs : (rank + 1) * local_bs]
+        ys = y[rank * local_bs : (rank + 1) * local_bs]
+
+        opt.zero_grad(set_to_none=True)
+        loss = loss_fn(ddp_net(xs), ys)
+        loss.backward()
+        opt.step()
+
+        gold_opt.zero_grad(set_to_none=True)
+        loss_fn(gold(x), y).backward()
+        gold_opt.step()
+
+    for (n1, p1), (n2, p2) in zip(
+        ddp_net.module.named_parameters(), gold.named_parameters()
+    ):
+        assert torch.allclose(p1, p2, atol=1e-5), f"{n1} diverged from golden"
+    # running stats must equal global-batch stats too
+    for (n1, b1), (n2, b2) in zip(
+        ddp_net.module.named_buffers(), gold.named_buffers()
+    ):
+        assert torch.allclose(b1.float(), b2.float(), atol=1e-5), n1
+
+
+def test_golden_model_equivalence(tmp_path):
+    _spawn("_golden_body", tmp_path)
+
+
+# --------------------------------------------------------------------------
+def _uneven_running_stats_body(rank):
+    """Different per-rank batch sizes: running stats must equal the analytic
+    stats of the concatenated global data (≈ distributed_test.py:6065-6102)."""
+    import msbn
+    from msbn.nn import SyncBatchNorm
+
+    bn = SyncBatchNorm(4, momentum=None)  # cumulative average
+    bn.train()
+    n_local = 2 if rank == 0 else 6
+    steps = 20
+    datas = []
+    for it in range(steps):
+        full = []
+        for r in range(WORLD):
+            g = torch.Generator().manual_seed(1000 + it * WORLD + r)
+            full.append(torch.randn(2 if r == 0 else 6, 4, 3, 3, generator=g))
+        datas.append(full)
+        bn(full[rank])
+    allx = torch.cat([torch.cat(f, dim=0) for f in datas], dim=0).double()
+    ref_mean = allx.mean(dim=(0, 2, 3))
+    # cumulative running var averages per-step unbiased global vars
+    per_step_vars = [
+        torch.cat(f, dim=0).double().var(dim=(0, 2, 3), unbiased=True)
+        for f in datas
+    ]
+    ref_var = torch.stack(per_step_vars).mean(0)
+    assert torch.allclose(bn.running_mean.double(), ref_mean, atol=1e-3)
+    assert torch.allclose(bn.running_var.double(), ref_var, atol=2e-2)
+
+
+def test_uneven_input_sizes_running_stats(tmp_path):
+    _spawn("_uneven_running_stats_body", tmp_path)
+
+
+# --------------------------------------------------------------------------
+def _uneven_grad_body(rank):
+    """Counts-weighted backward with different per-rank batches: grads of
+    SyncBN on shards == grads of plain BN on the concatenation
+    (≈ distributed_test.py:6109-6126)."""
+    import msbn
+    from msbn.nn import SyncBatchNorm
+
+    torch.manual_seed(5)
+    bs = rank + 2
+    xs = [torch.randn(r + 2, 3, 4, 4, dtype=torch.float64).float()
+          for r in range(WORLD)]
+    x_local = xs[rank].clone().requires_grad_(True)
+
+    bn = SyncBatchNorm(3)
+    y = bn(x_local)
+    y.sum().backward()
+
+    # oracle: plain BN over the concatenation
+    x_all = torch.cat(xs, dim=0).clone().requires_grad_(True)
+    tbn = torch.nn.BatchNorm2d(3)
+    with torch.no_grad():
+        tbn.weight.copy_(bn.weight)
+        tbn.bias.copy_(bn.bias)
+    ty = tbn(x_all)
+    ty.sum().backward()
+    off = sum(r + 2 for r in range(rank))
+    ref_slice = x_all.grad[off : off + bs]
+    assert torch.allclose(x_local.grad, ref_slice, atol=1e-4)
+
+
+def test_uneven_input_sizes_gradient(tmp_path):
+    _spawn("_uneven_grad_body", tmp_path)
+
+
+# --------------------------------------------------------------------------
+def _no_sync_body(rank):
+    import msbn
+
+    torch.manual_seed(0)
+    net = msbn.parallel.DistributedDataParallel(torch.nn.Linear(4, 2))
+    x = torch.full((2, 4), float(rank + 1))
+    with net.no_sync():
+        net(x).sum().backward()  # local grads only, accumulated
+    net(x).sum().backward()  # synced step: grads averaged INCL. accumulation
+    # grad of w wrt sum over batch = sum of x rows; two backwards accumulate
+    local = 2 * x.sum(0)  # per-rank accumulated (2 backwards)
+    expect = (2 * torch.full((4,), 1.0 * 2) + 2 * torch.full((4,), 2.0 * 2)) / 2
+    assert torch.allclose(net.module.weight.grad[0], expect), (
+        net.module.weight.grad[0],
+        expect,
+        local,
+    )
+
+
+def test_no_sync_accumulation(tmp_path):
+    _spawn("_no_sync_body", tmp_path)
+
+
+# --------------------------------------------------------------------------
+def _empty_input_body(rank):
+    """Rank 1 feeds an empty batch; SyncBN must not deadlock and stats must
+    come from rank 0 only (stock zero-count semantics, _functions.py:50-57)."""
+    import msbn
+    from msbn.nn import SyncBatchNorm
+
+    bn = SyncBatchNorm(3)
+    bn.train()
+    if rank == 0:
+        x = torch.randn(4, 3, 2, 2, requires_grad=True)
+    else:
+        x = torch.randn(0, 3, 2, 2, requires_grad=True)
+    y = bn(x)
+    y.sum().backward()
+    if rank == 0:
+        ref = torch.nn.functional.batch_norm(
+            x.detach(), None, None, bn.weight.detach(), bn.bias.detach(),
+            training=True, eps=bn.eps,
+        )
+        assert torch.allclose(y, ref, atol=1e-5)
+    else:
+        assert y.shape == x.shape
+
+
+def test_empty_input_rank(tmp_path):
+    _spawn("_empty_input_body", tmp_path)
+
+
+# --------------------------------------------------------------------------
+def _checkpoint_body(rank):
+    import msbn
+
+    torch.manual_seed(1)
+    net = msbn.convert_sync_batchnorm(msbn.models.SimpleCNN(width=8))
+    net = msbn.parallel.DistributedDataParallel(net)
+    x = torch.randn(2, 3, 8, 8)
+    net(x).sum().backward()
+    path = os.environ.get("MSBN_TEST_CKPT", "/tmp/msbn_ckpt_test.pt")
+    msbn.utils.save_checkpoint(path, net, epoch=3)
+    net2 = msbn.convert_sync_batchnorm(msbn.models.SimpleCNN(width=8))
+    net2 = msbn.parallel.DistributedDataParallel(net2)
+    epoch = msbn.utils.load_checkpoint(path, net2)
+    assert epoch == 3
+    for p1, p2 in zip(net.module.parameters(), net2.module.parameters()):
+        assert torch.equal(p1, p2)
+
+
+def test_checkpoint_roundtrip(tmp_path):
+    os.environ["MSBN_TEST_CKPT"] = str(tmp_path / "ckpt.pt")
+    _spawn("_checkpoint_body", tmp_path)
+
+
+# --------------------------------------------------------------------------
+def _rebuild_buckets_body(rank):
+    import msbn
+
+    torch.manual_seed(2)
+    net = msbn.parallel.DistributedDataParallel(
+        torch.nn.Sequential(torch.nn.Linear(8, 32), torch.nn.ReLU(),
+                            torch.nn.Linear(32, 4))
+    )
+    opt = torch.optim.SGD(net.parameters(), lr=0.01)
+    for it in range(3):
+        opt.zero_grad(set_to_none=True)
+        net(torch.randn(4, 8)).sum().backward()
+        opt.step()
+    assert net.reducer.rebuilt()
+    # all ranks hold identical bucket binning after the broadcast
+    import msbn._C  # noqa
+
+    idx = net.reducer.get_bucket_indices()
+    t = torch.tensor([i for b in idx for i in b], dtype=torch.long)
+    t0 = t.clone()
+    dist.broadcast(t0, src=0)
+    assert torch.equal(t, t0)
+
+
+def test_rebuild_buckets_consistent(tmp_path):
+    _spawn("_rebuild_buckets_body", tmp_path)

@@ -1,0 +1,96 @@
+"""End-to-end launcher test: the reference command line (README.md:98-100)
+through msbn.launch / msbn.run on CPU/gloo (BASELINE.json config 1)."""
+
+import os
+import subprocess
+import sys
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+WORKER = r"""
+import argparse, os, sys
+sys.path.insert(0, {repo!r})
+import torch, torch.distributed as dist
+import msbn
+
+# Step 1 (README.md:15-19): the --local_rank contract
+parser = argparse.ArgumentParser()
+parser.add_argument('--local_rank', '--local-rank', type=int,
+                    default=int(os.environ.get('LOCAL_RANK', 0)),
+                    dest='local_rank')
+parser.add_argument('--ngpu', type=int, default=2)
+args = parser.parse_args()
+
+# Step 2 (README.md:26-36): init_process_group env:// (gloo here: no GPU)
+dist.init_process_group('gloo', init_method='env://',
+                        world_size=args.ngpu, rank=args.local_rank)
+
+# Steps 3-5: convert + DDP + sampler
+torch.manual_seed(0)
+net = msbn.nn.SyncBatchNorm.convert_sync_batchnorm(msbn.models.SimpleCNN(width=8))
+net = msbn.parallel.DistributedDataParallel(net)
+ds = msbn.data.SyntheticImageDataset(length=16, shape=(3, 16, 16), num_classes=10)
+sampler = msbn.data.DistributedSampler(ds, num_replicas=args.ngpu,
+                                       rank=args.local_rank)
+loader = torch.utils.data.DataLoader(ds, batch_size=4, sampler=sampler,
+                                     drop_last=True, num_workers=0)
+opt = torch.optim.SGD(net.parameters(), lr=0.01)
+loss_fn = torch.nn.CrossEntropyLoss()
+for epoch in range(2):
+    sampler.set_epoch(epoch)
+    for x, y in loader:
+        opt.zero_grad()
+        loss = loss_fn(net(x), y)
+        loss.backward()
+        opt.step()
+if args.local_rank == 0:
+    print('LAUNCH_OK', loss.item())
+"""
+
+
+def _run_launcher(module, extra_args, tmp_path):
+    script = tmp_path / "worker.py"
+    script.write_text(WORKER.format(repo=REPO))
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO
+    cmd = [
+        sys.executable, "-m", module, "--nproc_per_node=2",
+        "--master-addr", "127.0.0.1",
+    ] + extra_args + [str(script), "--ngpu=2"]
+    r = subprocess.run(cmd, capture_output=True, text=True, timeout=180,
+                       env=env, cwd=REPO)
+    assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
+    assert "LAUNCH_OK" in r.stdout
+
+
+def test_msbn_launch_argv_contract(tmp_path):
+    """legacy launch: --local-rank=<r> injected on argv."""
+    _run_launcher("msbn.launch", [], tmp_path)
+
+
+def test_msbn_run_env_contract(tmp_path):
+    """torchrun-style: LOCAL_RANK via env only."""
+    _run_launcher("msbn.run", [], tmp_path)
+
+
+def test_restart_on_failure(tmp_path):
+    script = tmp_path / "flaky.py"
+    marker = tmp_path / "marker"
+    script.write_text(
+        "import os, sys\n"
+        f"m = {str(marker)!r}\n"
+        "if not os.path.exists(m):\n"
+        "    open(m, 'w').close()\n"
+        "    if os.environ['RANK'] == '1':\n"
+        "        sys.exit(3)\n"
+        "print('SECOND_TRY_OK')\n"
+    )
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO
+    r = subprocess.run(
+        [sys.executable, "-m", "msbn.run", "--nproc_per_node=2",
+         "--max-restarts=1", str(script)],
+        capture_output=True, text=True, timeout=120, env=env, cwd=REPO,
+    )
+    assert r.returncode == 0, r.stderr
+    assert "SECOND_TRY_OK" in r.stdout
