@@ -60,16 +60,25 @@ def main():
     p.add_argument("--batch-size", type=int, default=256,
                    help="per-GPU batch (weak scaling)")
     p.add_argument("--model", type=str, default="resnet50")
-    p.add_argument("--memory-format", type=str, default="channels_last",
+    # NCHW default: measured faster than channels_last for bf16 MIOpen convs
+    # on gfx950 (profiles/r01_single_gpu.md)
+    p.add_argument("--memory-format", type=str, default="contiguous",
                    choices=["channels_last", "contiguous"])
     p.add_argument("--dtype", type=str, default="bf16",
                    choices=["bf16", "fp32"])
     p.add_argument("--stock", action="store_true",
                    help="use stock torch SyncBatchNorm+DDP (comparison line)")
+    p.add_argument("--no-benchmark", action="store_true",
+                   help="disable MIOpen conv autotune (cudnn.benchmark)")
+    p.add_argument("--graph", action="store_true",
+                   help="capture the train step in a hipGraph (single GPU)")
     p.add_argument("--local_rank", "--local-rank", type=int,
                    default=int(os.environ.get("LOCAL_RANK", 0)),
                    dest="local_rank")
     args = p.parse_args()
+
+    if not args.no_benchmark:
+        torch.backends.cudnn.benchmark = True  # MIOpen conv autotune
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
@@ -133,16 +142,42 @@ def main():
         opt.step()
         return loss
 
-    for _ in range(args.warmup):
-        step()
+    graph = None
+    if args.graph and use_cuda and not distributed:
+        # hipGraph-captured whole train step (fwd+bwd+optimizer): kills the
+        # ~500 per-step kernel-launch round trips.  msbn's BN ops are
+        # capture-safe by design (no host syncs; zero-count masking is
+        # in-kernel).
+        for _ in range(max(args.warmup, 3)):
+            step()
+        opt.zero_grad(set_to_none=False)
+        torch.cuda.synchronize()
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            out = model(x)
+            loss = loss_fn(out.float(), y)
+            loss.backward()
+            opt.step()
+            # zero grads IN-GRAPH so replays are self-contained
+            torch._foreach_zero_([p.grad for p in model.parameters()
+                                  if p.grad is not None])
+        for _ in range(args.warmup):
+            graph.replay()
+    else:
+        for _ in range(args.warmup):
+            step()
 
     if distributed:
         dist.barrier()
     if use_cuda:
         torch.cuda.synchronize()
     t0 = time.perf_counter()
-    for _ in range(args.steps):
-        step()
+    if graph is not None:
+        for _ in range(args.steps):
+            graph.replay()
+    else:
+        for _ in range(args.steps):
+            step()
     if use_cuda:
         torch.cuda.synchronize()
     if distributed:
@@ -182,6 +217,8 @@ def main():
                 "memory_format": args.memory_format,
                 "parallelism": f"dp{n_gpus}",
                 "impl": "stock" if args.stock else "msbn",
+                "hip_graph": graph is not None,
+                "conv_autotune": not args.no_benchmark,
             },
         }))
 
