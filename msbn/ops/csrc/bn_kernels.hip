@@ -69,7 +69,35 @@ __global__ void bn_stats_partial_nchw(const T* __restrict__ x,
   block_reduce_pair(a, b, lds);
   if (threadIdx.x == 0) {
     const int64_t chunk = (int64_t)blockIdx.y * gridDim.z + blockIdx.z;
-    double* out = ws + (chunk * C + c) * 2;
+    double* out = ws + (c * gridDim.y * gridDim.z + chunk) * 2;
+    out[0] = a;
+    out[1] = b;
+  }
+}
+
+// V==1 fallback for layouts where the spatial extent is small or odd
+// (e.g. S=49 at C=2048): flat index over the whole (n, s) chunk keeps every
+// lane busy (the row-loop form idles blockDim-S threads per row).
+template <typename T>
+__global__ void bn_stats_partial_nchw_flat(const T* __restrict__ x,
+                                           double* __restrict__ ws, int64_t N,
+                                           int64_t C, int64_t S,
+                                           int64_t chunk_len) {
+  const int64_t c = blockIdx.x;
+  const int64_t NS = N * S;
+  const int64_t p0 = (int64_t)blockIdx.y * chunk_len;
+  const int64_t p1 = i64min(p0 + chunk_len, NS);
+  double a = 0.0, b = 0.0;
+  for (int64_t p = p0 + threadIdx.x; p < p1; p += blockDim.x) {
+    const int64_t n = p / S, s = p - n * S;
+    const float v = to_f(x[(n * C + c) * S + s]);
+    a += v;
+    b += (double)v * v;
+  }
+  __shared__ double lds[2 * (MSBN_BLOCK / MSBN_WAVE)];
+  block_reduce_pair(a, b, lds);
+  if (threadIdx.x == 0) {
+    double* out = ws + (c * gridDim.y + blockIdx.y) * 2;
     out[0] = a;
     out[1] = b;
   }
@@ -133,7 +161,7 @@ __global__ void bn_stats_partial_nhwc(const T* __restrict__ x,
 #pragma unroll
     for (int k = 0; k < V; ++k) {
       if (c + k < C) {
-        double* out = ws + (chunk * C + c + k) * 2;
+        double* out = ws + ((c + k) * gridDim.y + chunk) * 2;
         out[0] = sdata[(threadIdx.x * V + k) * 2];
         out[1] = sdata[(threadIdx.x * V + k) * 2 + 1];
       }
@@ -142,7 +170,10 @@ __global__ void bn_stats_partial_nhwc(const T* __restrict__ x,
 }
 
 // =====================================================================
-// stats finalize: combine chunk partials -> mean/invstd (+count, +running)
+// stats finalize: combine chunk partials -> mean/invstd (+count, +running).
+// ONE WAVE per channel (ws is channel-major [C][nchunks][2] -> coalesced
+// lane-strided loads + wave64 shuffle reduce); nchunks may be in the
+// thousands for small-C layers without serializing.
 // =====================================================================
 template <typename RT>
 __global__ void bn_stats_finalize(const double* __restrict__ ws, int nchunks,
@@ -152,13 +183,17 @@ __global__ void bn_stats_finalize(const double* __restrict__ ws, int nchunks,
                                   float* __restrict__ count_out,
                                   RT* __restrict__ rmean, RT* __restrict__ rvar,
                                   float momentum) {
-  const int64_t c = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int lane = threadIdx.x & (MSBN_WAVE - 1);
+  const int wid = threadIdx.x >> 6;
+  const int64_t c = (int64_t)blockIdx.x * (blockDim.x >> 6) + wid;
   if (c >= C) return;
   double a = 0.0, b = 0.0;
-  for (int ch = 0; ch < nchunks; ++ch) {
-    a += ws[((int64_t)ch * C + c) * 2];
-    b += ws[((int64_t)ch * C + c) * 2 + 1];
+  for (int ch = lane; ch < nchunks; ch += MSBN_WAVE) {
+    a += ws[(c * nchunks + ch) * 2];
+    b += ws[(c * nchunks + ch) * 2 + 1];
   }
+  wave_reduce_pair(a, b);
+  if (lane != 0) return;
   const double m = a / count;
   double var = b / count - m * m;
   var = var > 0.0 ? var : 0.0;
@@ -361,7 +396,34 @@ __global__ void bn_bwd_reduce_partial_nchw(const T* __restrict__ dy,
   block_reduce_pair(a, b, lds);
   if (threadIdx.x == 0) {
     const int64_t chunk = (int64_t)blockIdx.y * gridDim.z + blockIdx.z;
-    double* out = ws + (chunk * C + c) * 2;
+    double* out = ws + (c * gridDim.y * gridDim.z + chunk) * 2;
+    out[0] = a;
+    out[1] = b;
+  }
+}
+
+template <typename T>
+__global__ void bn_bwd_reduce_partial_nchw_flat(
+    const T* __restrict__ dy, const T* __restrict__ x,
+    const float* __restrict__ mean, double* __restrict__ ws, int64_t N,
+    int64_t C, int64_t S, int64_t chunk_len) {
+  const int64_t c = blockIdx.x;
+  const float m = mean[c];
+  const int64_t NS = N * S;
+  const int64_t p0 = (int64_t)blockIdx.y * chunk_len;
+  const int64_t p1 = i64min(p0 + chunk_len, NS);
+  double a = 0.0, b = 0.0;
+  for (int64_t p = p0 + threadIdx.x; p < p1; p += blockDim.x) {
+    const int64_t n = p / S, s = p - n * S;
+    const int64_t e = (n * C + c) * S + s;
+    const float g = to_f(dy[e]);
+    a += g;
+    b += (double)g * (to_f(x[e]) - m);
+  }
+  __shared__ double lds[2 * (MSBN_BLOCK / MSBN_WAVE)];
+  block_reduce_pair(a, b, lds);
+  if (threadIdx.x == 0) {
+    double* out = ws + (c * gridDim.y + blockIdx.y) * 2;
     out[0] = a;
     out[1] = b;
   }
@@ -427,7 +489,7 @@ __global__ void bn_bwd_reduce_partial_nhwc(const T* __restrict__ dy,
 #pragma unroll
     for (int k = 0; k < V; ++k) {
       if (c + k < C) {
-        double* out = ws + (chunk * C + c + k) * 2;
+        double* out = ws + ((c + k) * gridDim.y + chunk) * 2;
         out[0] = sdata[(threadIdx.x * V + k) * 2];
         out[1] = sdata[(threadIdx.x * V + k) * 2 + 1];
       }
@@ -435,6 +497,7 @@ __global__ void bn_bwd_reduce_partial_nhwc(const T* __restrict__ dy,
   }
 }
 
+// one wave per channel (see bn_stats_finalize)
 template <typename WT>
 __global__ void bn_bwd_reduce_finalize(const double* __restrict__ ws,
                                        int nchunks, int64_t C,
@@ -443,13 +506,17 @@ __global__ void bn_bwd_reduce_finalize(const double* __restrict__ ws,
                                        float* __restrict__ sum_dy_xmu,
                                        WT* __restrict__ grad_weight,
                                        WT* __restrict__ grad_bias) {
-  const int64_t c = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int lane = threadIdx.x & (MSBN_WAVE - 1);
+  const int wid = threadIdx.x >> 6;
+  const int64_t c = (int64_t)blockIdx.x * (blockDim.x >> 6) + wid;
   if (c >= C) return;
   double a = 0.0, b = 0.0;
-  for (int ch = 0; ch < nchunks; ++ch) {
-    a += ws[((int64_t)ch * C + c) * 2];
-    b += ws[((int64_t)ch * C + c) * 2 + 1];
+  for (int ch = lane; ch < nchunks; ch += MSBN_WAVE) {
+    a += ws[(c * nchunks + ch) * 2];
+    b += ws[(c * nchunks + ch) * 2 + 1];
   }
+  wave_reduce_pair(a, b);
+  if (lane != 0) return;
   if (sum_dy != nullptr) sum_dy[c] = (float)a;
   if (sum_dy_xmu != nullptr) sum_dy_xmu[c] = (float)b;
   if (grad_weight != nullptr) grad_weight[c] = from_f<WT>((float)(b * invstd[c]));
@@ -547,6 +614,30 @@ Layout get_layout(const at::Tensor& t) {
 }
 
 inline int64_t cdiv(int64_t a, int64_t b) { return (a + b - 1) / b; }
+inline int64_t clamp64(int64_t v, int64_t lo, int64_t hi) {
+  return v < lo ? lo : (v > hi ? hi : v);
+}
+
+// channels per finalize block (one wave each)
+constexpr int kFinalizeWavesPerBlock = MSBN_BLOCK / MSBN_WAVE;
+
+struct FlatGrid {
+  dim3 grid;
+  int64_t chunk_len;
+  int nchunks;
+};
+
+// grid for the V==1 flat NCHW partial kernels: (C, nchunks)
+FlatGrid flat_grid(int64_t C, int64_t NS, int64_t target_blocks) {
+  FlatGrid g{};
+  int64_t nchunks =
+      clamp64(target_blocks / std::max<int64_t>(C, 1), 1, cdiv(NS, 1024));
+  g.chunk_len = cdiv(NS, nchunks);
+  nchunks = cdiv(NS, g.chunk_len);
+  g.grid = dim3((unsigned)C, (unsigned)nchunks);
+  g.nchunks = (int)nchunks;
+  return g;
+}
 
 // pick vector width: 16B/lane when layout & alignment allow
 template <typename T>
@@ -719,6 +810,15 @@ void stats_into(const at::Tensor& input, double eps, float* mean_p,
     int nchunks = 0;
     if (!L.nhwc) {
       const int v = pick_v<native_t>(x, nullptr, nullptr, L.S);
+      if (v == 1) {
+        auto g = flat_grid(L.C, L.rows, kTargetBlocks);
+        nchunks = g.nchunks;
+        ws = at::empty({(int64_t)nchunks * L.C * 2},
+                       input.options().dtype(at::kDouble));
+        hipLaunchKernelGGL((bn_stats_partial_nchw_flat<native_t>), g.grid,
+                           dim3(MSBN_BLOCK), 0, stream, x,
+                           ws.data_ptr<double>(), L.N, L.C, L.S, g.chunk_len);
+      } else {
       auto g = nchw_grid(L.N, L.C, L.S, v);
       nchunks = g.nchunks;
       ws = at::empty({(int64_t)nchunks * L.C * 2},
@@ -729,6 +829,7 @@ void stats_into(const at::Tensor& input, double eps, float* mean_p,
                            ws.data_ptr<double>(), L.N, L.C, L.S, g.chunkN,
                            g.chunkS);
       });
+      }
     } else {
       const int v = pick_v<native_t>(x, nullptr, nullptr, L.C);
       auto g = nhwc_grid(L.rows, L.C, v);
@@ -742,7 +843,7 @@ void stats_into(const at::Tensor& input, double eps, float* mean_p,
                            g.lpr);
       });
     }
-    const int fgrid = (int)cdiv(L.C, MSBN_BLOCK);
+    const int fgrid = (int)cdiv(L.C, kFinalizeWavesPerBlock);
     MSBN_DISPATCH_RSTAT(rstat_type, "batch_norm_stats", [&] {
       rstat_t* rm = rmean ? reinterpret_cast<rstat_t*>(rmean->data_ptr())
                           : nullptr;
@@ -931,6 +1032,16 @@ batch_norm_backward_reduce(const at::Tensor& grad_out, const at::Tensor& input,
     int nchunks = 0;
     if (!L.nhwc) {
       const int v = pick_v<native_t>(x, dy, nullptr, L.S);
+      if (v == 1) {
+        auto g = flat_grid(L.C, L.rows, kTargetBlocks);
+        nchunks = g.nchunks;
+        ws = at::empty({(int64_t)nchunks * L.C * 2},
+                       input.options().dtype(at::kDouble));
+        hipLaunchKernelGGL((bn_bwd_reduce_partial_nchw_flat<native_t>), g.grid,
+                           dim3(MSBN_BLOCK), 0, stream, dy, x,
+                           mean.data_ptr<float>(), ws.data_ptr<double>(), L.N,
+                           L.C, L.S, g.chunk_len);
+      } else {
       auto g = nchw_grid(L.N, L.C, L.S, v);
       nchunks = g.nchunks;
       ws = at::empty({(int64_t)nchunks * L.C * 2},
@@ -941,6 +1052,7 @@ batch_norm_backward_reduce(const at::Tensor& grad_out, const at::Tensor& input,
                            mean.data_ptr<float>(), ws.data_ptr<double>(), L.N,
                            L.C, L.S, g.chunkN, g.chunkS);
       });
+      }
     } else {
       const int v = pick_v<native_t>(x, dy, nullptr, L.C);
       auto g = nhwc_grid(L.rows, L.C, v);
@@ -954,7 +1066,7 @@ batch_norm_backward_reduce(const at::Tensor& grad_out, const at::Tensor& input,
                            L.rows, L.C, g.chunk_rows, g.lpr);
       });
     }
-    const int fgrid = (int)cdiv(L.C, MSBN_BLOCK);
+    const int fgrid = (int)cdiv(L.C, kFinalizeWavesPerBlock);
     MSBN_DISPATCH_RSTAT(wtype, "bn_bwd_reduce", [&] {
       rstat_t* gw = weight_g ? reinterpret_cast<rstat_t*>(grad_weight.data_ptr())
                              : nullptr;
