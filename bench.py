@@ -60,9 +60,10 @@ def main():
     p.add_argument("--batch-size", type=int, default=256,
                    help="per-GPU batch (weak scaling)")
     p.add_argument("--model", type=str, default="resnet50")
-    # NCHW default: measured faster than channels_last for bf16 MIOpen convs
-    # on gfx950 (profiles/r01_single_gpu.md)
-    p.add_argument("--memory-format", type=str, default="contiguous",
+    # channels_last default: fastest measured path on gfx950 (MIOpen igemm
+    # kernels are NHWC-native -> no transposes; msbn NHWC BN kernels are
+    # C-vectorized). See profiles/r01_single_gpu.md.
+    p.add_argument("--memory-format", type=str, default="channels_last",
                    choices=["channels_last", "contiguous"])
     p.add_argument("--dtype", type=str, default="bf16",
                    choices=["bf16", "fp32"])
