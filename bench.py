@@ -35,11 +35,11 @@ def cast_bf16_keep_bn_fp32(model):
     return model
 
 
-def build_model(name, memory_format, dtype, device, sync):
+def build_model(name, memory_format, dtype, device, sync, fused=True):
     if name == "resnet50":
-        model = msbn.models.resnet50()
+        model = msbn.models.resnet50(fused=fused)
     elif name == "resnet18":
-        model = msbn.models.resnet18()
+        model = msbn.models.resnet18(fused=fused)
     else:
         raise ValueError(name)
     if sync:
@@ -74,6 +74,8 @@ def main():
                         "faster steps but minutes of one-time find cost")
     p.add_argument("--graph", action="store_true",
                    help="capture the train step in a hipGraph (single GPU)")
+    p.add_argument("--no-fused", action="store_true",
+                   help="disable the fused BN(+add)+ReLU epilogue modules")
     p.add_argument("--local_rank", "--local-rank", type=int,
                    default=int(os.environ.get("LOCAL_RANK", 0)),
                    dest="local_rank")
@@ -116,7 +118,8 @@ def main():
                 output_device=args.local_rank if use_cuda else None,
             )
     else:
-        model = build_model(args.model, mem_fmt, dtype, device, sync=True)
+        model = build_model(args.model, mem_fmt, dtype, device, sync=True,
+                            fused=not args.no_fused)
         if distributed:
             model = msbn.parallel.DistributedDataParallel(
                 model,
@@ -221,6 +224,7 @@ def main():
                 "impl": "stock" if args.stock else "msbn",
                 "hip_graph": graph is not None,
                 "conv_autotune": args.benchmark,
+                "fused_bn_act": not (args.stock or args.no_fused),
             },
         }))
 

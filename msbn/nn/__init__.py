@@ -6,6 +6,7 @@ from msbn.nn.batchnorm import (  # noqa: F401
     convert_sync_batchnorm,
 )
 from msbn.nn.functions import SyncBatchNormFunction  # noqa: F401
+from msbn.nn.fused import SyncBatchNormAct2d, SyncBatchNormActFunction  # noqa: F401
 
 __all__ = [
     "BatchNorm1d",

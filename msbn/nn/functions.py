@@ -38,6 +38,71 @@ def _is_nccl_like(process_group) -> bool:
         return False
 
 
+def _contig(t: torch.Tensor) -> torch.Tensor:
+    if t.is_contiguous(memory_format=torch.channels_last) or t.is_contiguous(
+        memory_format=torch.channels_last_3d
+    ):
+        return t
+    return t.contiguous()
+
+
+def compute_sync_stats(
+    input: torch.Tensor,
+    eps: float,
+    momentum: float,
+    running_mean: Optional[torch.Tensor],
+    running_var: Optional[torch.Tensor],
+    process_group,
+    world_size: int,
+):
+    """Local packed moments -> all_gather over RCCL/xGMI (S5) -> counts-weighted
+    combine with in-kernel zero-count masking + running-stats update.
+    Returns (mean, invstd, count_sum[1])."""
+    C = int(input.shape[1])
+    local_count = input.numel() // C if C > 0 else 0
+
+    if world_size > 1:
+        packed = torch.empty(2 * C + 1, dtype=torch.float32, device=input.device)
+        if local_count > 0:
+            ops.batch_norm_stats_packed(input, eps, packed)
+        else:
+            packed.zero_()
+
+        packed_all = torch.empty(
+            (world_size, 2 * C + 1), dtype=torch.float32, device=input.device
+        )
+        if _is_nccl_like(process_group):
+            dist.all_gather_into_tensor(packed_all, packed, group=process_group)
+        else:
+            chunks = list(packed_all.unbind(0))
+            dist.all_gather(chunks, packed, group=process_group)
+            packed_all = torch.stack(chunks, dim=0)
+
+        mean, invstd, count_sum = ops.batch_norm_gather_stats_packed(
+            input, packed_all, running_mean, running_var, momentum, eps
+        )
+    else:
+        mean, invstd = ops.batch_norm_stats(input, eps)
+        count_sum = torch.full(
+            (1,), float(local_count), dtype=torch.float32, device=input.device
+        )
+        if running_mean is not None and local_count > 0:
+            with torch.no_grad():
+                var = invstd.to(torch.float32).pow(-2) - eps
+                unbiased = (
+                    var * (local_count / (local_count - 1.0))
+                    if local_count > 1
+                    else var
+                )
+                running_mean.mul_(1 - momentum).add_(
+                    mean.to(running_mean.dtype), alpha=momentum
+                )
+                running_var.mul_(1 - momentum).add_(
+                    unbiased.to(running_var.dtype), alpha=momentum
+                )
+    return mean, invstd, count_sum
+
+
 class SyncBatchNormFunction(torch.autograd.Function):
     @staticmethod
     def forward(
@@ -52,59 +117,17 @@ class SyncBatchNormFunction(torch.autograd.Function):
         process_group,
         world_size: int,
     ):
-        if not (
-            input.is_contiguous(memory_format=torch.channels_last)
-            or input.is_contiguous(memory_format=torch.channels_last_3d)
-        ):
-            input = input.contiguous()
+        input = _contig(input)
         if weight is not None:
             weight = weight.contiguous()
 
         C = int(input.shape[1])
         local_count = input.numel() // C if C > 0 else 0
 
-        if world_size > 1:
-            # -- packed local stats ---------------------------------------
-            packed = torch.empty(2 * C + 1, dtype=torch.float32, device=input.device)
-            if local_count > 0:
-                ops.batch_norm_stats_packed(input, eps, packed)
-            else:
-                packed.zero_()
-
-            # -- cross-replica gather over RCCL/xGMI (S5) ------------------
-            packed_all = torch.empty(
-                (world_size, 2 * C + 1), dtype=torch.float32, device=input.device
-            )
-            if _is_nccl_like(process_group):
-                dist.all_gather_into_tensor(packed_all, packed, group=process_group)
-            else:
-                chunks = list(packed_all.unbind(0))
-                dist.all_gather(chunks, packed, group=process_group)
-                packed_all = torch.stack(chunks, dim=0)
-
-            # -- combine + running-stats update (in-kernel masking) --------
-            mean, invstd, count_sum = ops.batch_norm_gather_stats_packed(
-                input, packed_all, running_mean, running_var, momentum, eps
-            )
-        else:
-            mean, invstd = ops.batch_norm_stats(input, eps)
-            count_sum = torch.full(
-                (1,), float(local_count), dtype=torch.float32, device=input.device
-            )
-            if running_mean is not None and local_count > 0:
-                with torch.no_grad():
-                    var = invstd.to(torch.float32).pow(-2) - eps
-                    unbiased = (
-                        var * (local_count / (local_count - 1.0))
-                        if local_count > 1
-                        else var
-                    )
-                    running_mean.mul_(1 - momentum).add_(
-                        mean.to(running_mean.dtype), alpha=momentum
-                    )
-                    running_var.mul_(1 - momentum).add_(
-                        unbiased.to(running_var.dtype), alpha=momentum
-                    )
+        mean, invstd, count_sum = compute_sync_stats(
+            input, eps, momentum, running_mean, running_var,
+            process_group, world_size,
+        )
 
         ctx.save_for_backward(input, weight, mean, invstd, count_sum)
         ctx.process_group = process_group

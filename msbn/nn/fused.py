@@ -1,0 +1,163 @@
+"""Fused SyncBatchNorm(+residual add)(+ReLU) — the msbn epilogue-fusion path.
+
+Mathematically identical to ``relu(bn(x) + residual)`` composed from separate
+modules, but executed as ONE elementwise kernel forward and mask-recomputing
+kernels backward (no clamp / add / threshold_backward kernels, no stored mask,
+~3x fewer full-tensor round trips on the BN backward — see
+profiles/r01_single_gpu.md for the measured eager-kernel cost this removes).
+
+``SyncBatchNormAct2d`` is used directly by the fused model variants
+(msbn.models.resnet{18,50}(fused=True)); ``convert_sync_batchnorm`` leaves it
+alone (it already IS a SyncBatchNorm subclass).
+"""
+
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+from msbn import ops
+from msbn.nn.batchnorm import SyncBatchNorm, _momentum_factor
+from msbn.nn.functions import _contig, compute_sync_stats
+
+
+class SyncBatchNormActFunction(torch.autograd.Function):
+    @staticmethod
+    def forward(
+        ctx,
+        input: torch.Tensor,
+        residual: Optional[torch.Tensor],
+        weight: Optional[torch.Tensor],
+        bias: Optional[torch.Tensor],
+        running_mean: Optional[torch.Tensor],
+        running_var: Optional[torch.Tensor],
+        eps: float,
+        momentum: float,
+        process_group,
+        world_size: int,
+        relu: bool,
+    ):
+        input = _contig(input)
+        if residual is not None:
+            residual = _contig(residual)
+        if weight is not None:
+            weight = weight.contiguous()
+        if bias is not None:
+            bias = bias.contiguous()
+
+        mean, invstd, count_sum = compute_sync_stats(
+            input, eps, momentum, running_mean, running_var,
+            process_group, world_size,
+        )
+        ctx.save_for_backward(input, residual, weight, bias, mean, invstd,
+                              count_sum)
+        ctx.relu = relu
+        ctx.process_group = process_group
+        ctx.world_size = world_size
+        return ops.batch_norm_elemt_act(
+            input, residual, weight, bias, mean, invstd, relu
+        )
+
+    @staticmethod
+    def backward(ctx, grad_output: torch.Tensor):
+        grad_output = _contig(grad_output)
+        input, residual, weight, bias, mean, invstd, count_sum = (
+            ctx.saved_tensors
+        )
+        relu = ctx.relu
+        process_group = ctx.process_group
+        world_size = ctx.world_size
+        need_input_g = ctx.needs_input_grad[0]
+        need_res_g = residual is not None and ctx.needs_input_grad[1]
+        need_weight_g = weight is not None and ctx.needs_input_grad[2]
+        need_bias_g = bias is not None and ctx.needs_input_grad[3]
+
+        C = int(input.shape[1])
+        sum_dy, sum_dy_xmu, grad_weight, grad_bias = (
+            ops.batch_norm_backward_reduce_act(
+                grad_output, input, residual, mean, invstd, weight, bias,
+                relu, need_input_g, need_weight_g, need_bias_g,
+            )
+        )
+        grad_input = grad_res = None
+        if need_input_g or need_res_g:
+            if world_size > 1:
+                combined = torch.cat([sum_dy, sum_dy_xmu])
+                dist.all_reduce(combined, dist.ReduceOp.SUM, group=process_group)
+                sum_dy, sum_dy_xmu = combined[:C], combined[C:]
+            grad_input, grad_res = ops.batch_norm_backward_elemt_act(
+                grad_output, input, residual, mean, invstd, weight, bias,
+                sum_dy, sum_dy_xmu, count_sum, relu, need_res_g,
+            )
+        return (
+            grad_input if need_input_g else None,
+            grad_res,
+            grad_weight if need_weight_g else None,
+            grad_bias if need_bias_g else None,
+            None, None, None, None, None, None, None,
+        )
+
+
+class SyncBatchNormAct2d(SyncBatchNorm):
+    """SyncBatchNorm with a fused (optional residual-add +) ReLU epilogue.
+
+    forward(x, residual=None) == relu(bn(x) + residual) with identical math.
+    """
+
+    def __init__(self, num_features, eps=1e-5, momentum=0.1, affine=True,
+                 track_running_stats=True, process_group=None, device=None,
+                 dtype=None, relu: bool = True):
+        super().__init__(num_features, eps, momentum, affine,
+                         track_running_stats, process_group, device, dtype)
+        self.relu = relu
+
+    def forward(self, input: torch.Tensor,
+                residual: Optional[torch.Tensor] = None) -> torch.Tensor:
+        self._check_input_dim(input)
+        self._check_non_zero_input_channels(input)
+
+        bn_training = self.training or (
+            self.running_mean is None and self.running_var is None
+        )
+        if self.training and self.track_running_stats:
+            if self.num_batches_tracked is not None:
+                self.num_batches_tracked.add_(1)
+        factor = _momentum_factor(self) if self.training else (
+            self.momentum if self.momentum is not None else 0.0
+        )
+
+        if not bn_training:
+            # eval: running-stat normalize, composed (differentiable) path
+            rm = self.running_mean.to(torch.float32)
+            rv = self.running_var.to(torch.float32)
+            invstd = torch.rsqrt(rv + self.eps)
+            scale = invstd
+            shift = -rm * invstd
+            if self.weight is not None:
+                wf = self.weight.to(torch.float32)
+                scale = scale * wf
+                shift = shift * wf
+            if self.bias is not None:
+                shift = shift + self.bias.to(torch.float32)
+            shape = [1] * input.dim()
+            shape[1] = input.shape[1]
+            out = input.to(torch.float32) * scale.reshape(shape) + \
+                shift.reshape(shape)
+            if residual is not None:
+                out = out + residual.to(torch.float32)
+            if self.relu:
+                out = torch.relu(out)
+            return out.to(input.dtype)
+
+        process_group = None
+        world_size = 1
+        if dist.is_available() and dist.is_initialized():
+            process_group = self.process_group or dist.group.WORLD
+            world_size = dist.get_world_size(process_group)
+
+        running_mean = self.running_mean if self.track_running_stats else None
+        running_var = self.running_var if self.track_running_stats else None
+        return SyncBatchNormActFunction.apply(
+            input, residual, self.weight, self.bias, running_mean, running_var,
+            self.eps, factor, process_group, world_size, self.relu,
+        )

@@ -166,8 +166,51 @@ def batch_norm_backward_elemt(
     )
 
 
+def batch_norm_elemt_act(input, residual, weight, bias, mean, invstd,
+                         relu: bool):
+    if input.is_cuda:
+        return _require_hip().batch_norm_elemt_act(
+            input, residual, weight, bias, mean, invstd, relu
+        )
+    return _ref.batch_norm_elemt_act(
+        input, residual, weight, bias, mean, invstd, relu
+    )
+
+
+def batch_norm_backward_reduce_act(grad_out, input, residual, mean, invstd,
+                                   weight, bias, relu_mask, input_g, weight_g,
+                                   bias_g):
+    if input.is_cuda:
+        return _require_hip().batch_norm_backward_reduce_act(
+            grad_out, input, residual, mean, invstd, weight, bias, relu_mask,
+            input_g, weight_g, bias_g,
+        )
+    return _ref.batch_norm_backward_reduce_act(
+        grad_out, input, residual, mean, invstd, weight, bias, relu_mask,
+        input_g, weight_g, bias_g,
+    )
+
+
+def batch_norm_backward_elemt_act(grad_out, input, residual, mean, invstd,
+                                  weight, bias, sum_dy, sum_dy_xmu, count,
+                                  relu_mask, want_res_grad):
+    if input.is_cuda:
+        dx, dres = _require_hip().batch_norm_backward_elemt_act(
+            grad_out, input, residual, mean, invstd, weight, bias, sum_dy,
+            sum_dy_xmu, count, relu_mask, want_res_grad,
+        )
+        return dx, (dres if want_res_grad else None)
+    return _ref.batch_norm_backward_elemt_act(
+        grad_out, input, residual, mean, invstd, weight, bias, sum_dy,
+        sum_dy_xmu, count, relu_mask, want_res_grad,
+    )
+
+
 __all__ = [
     "hip_available",
+    "batch_norm_elemt_act",
+    "batch_norm_backward_reduce_act",
+    "batch_norm_backward_elemt_act",
     "batch_norm_stats",
     "batch_norm_stats_packed",
     "batch_norm_gather_stats_with_counts",

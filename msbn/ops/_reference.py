@@ -129,6 +129,69 @@ def batch_norm_elemt(
     return y.to(input.dtype)
 
 
+def _act_coefs(mean, invstd, weight, bias):
+    scale = invstd.to(torch.float32)
+    if weight is not None:
+        scale = scale * weight.to(torch.float32)
+    shift = -mean.to(torch.float32) * scale
+    if bias is not None:
+        shift = shift + bias.to(torch.float32)
+    return scale, shift
+
+
+def batch_norm_elemt_act(
+    input, residual, weight, bias, mean, invstd, relu: bool
+):
+    """y = relu?(x*scale + shift [+ residual]) — fused epilogue reference."""
+    scale, shift = _act_coefs(mean, invstd, weight, bias)
+    z = input.to(torch.float32) * _chan_view(scale, input) + _chan_view(
+        shift, input
+    )
+    if residual is not None:
+        z = z + residual.to(torch.float32)
+    if relu:
+        z = torch.relu(z)
+    return z.to(input.dtype)
+
+
+def _masked_grad(grad_out, input, residual, mean, invstd, weight, bias,
+                 relu_mask: bool):
+    g = grad_out.to(torch.float32)
+    if relu_mask:
+        scale, shift = _act_coefs(mean, invstd, weight, bias)
+        z = input.to(torch.float32) * _chan_view(scale, input) + _chan_view(
+            shift, input
+        )
+        if residual is not None:
+            z = z + residual.to(torch.float32)
+        g = torch.where(z > 0, g, torch.zeros_like(g))
+    return g
+
+
+def batch_norm_backward_reduce_act(
+    grad_out, input, residual, mean, invstd, weight, bias, relu_mask,
+    input_g, weight_g, bias_g,
+):
+    g = _masked_grad(grad_out, input, residual, mean, invstd, weight, bias,
+                     relu_mask)
+    return batch_norm_backward_reduce(
+        g, input, mean, invstd, weight, input_g, weight_g, bias_g
+    )
+
+
+def batch_norm_backward_elemt_act(
+    grad_out, input, residual, mean, invstd, weight, bias, sum_dy,
+    sum_dy_xmu, count, relu_mask, want_res_grad,
+):
+    g = _masked_grad(grad_out, input, residual, mean, invstd, weight, bias,
+                     relu_mask)
+    dx = batch_norm_backward_elemt(
+        g, input, mean, invstd, weight, sum_dy, sum_dy_xmu, count
+    )
+    dres = g.to(grad_out.dtype) if want_res_grad else None
+    return dx, dres
+
+
 def batch_norm_backward_reduce(
     grad_out: torch.Tensor,
     input: torch.Tensor,

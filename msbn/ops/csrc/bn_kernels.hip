@@ -303,10 +303,14 @@ __global__ void bn_affine_bwd(const float* __restrict__ mean,
 }
 
 // =====================================================================
-// elementwise: y = x*scale[c] + shift[c]
+// elementwise: y = act(x*scale[c] + shift[c] [+ res])
+// RELU / RES are compile-time: the fused BN(+add)+ReLU epilogue replaces the
+// eager clamp / add kernels and their full-tensor round trips (guide G13:
+// fuse into the producing kernel).
 // =====================================================================
-template <typename T, int V>
-__global__ void bn_elemt_nchw(const T* __restrict__ x, T* __restrict__ y,
+template <typename T, int V, bool RELU, bool RES>
+__global__ void bn_elemt_nchw(const T* __restrict__ x,
+                              const T* __restrict__ res, T* __restrict__ y,
                               const float* __restrict__ scale,
                               const float* __restrict__ shift, int64_t total,
                               int64_t C, int64_t S) {
@@ -317,19 +321,30 @@ __global__ void bn_elemt_nchw(const T* __restrict__ x, T* __restrict__ y,
     const int64_t c = (e / S) % C;
     const float sc = scale[c], sh = shift[c];
     if (V == 1) {
-      y[e] = from_f<T>(to_f(x[e]) * sc + sh);
+      float z = to_f(x[e]) * sc + sh;
+      if (RES) z += to_f(res[e]);
+      if (RELU) z = fmaxf(z, 0.f);
+      y[e] = from_f<T>(z);
     } else {
       Pack<T, V> px = *reinterpret_cast<const Pack<T, V>*>(&x[e]);
+      Pack<T, V> pr;
+      if (RES) pr = *reinterpret_cast<const Pack<T, V>*>(&res[e]);
       Pack<T, V> py;
 #pragma unroll
-      for (int k = 0; k < V; ++k) py.v[k] = from_f<T>(to_f(px.v[k]) * sc + sh);
+      for (int k = 0; k < V; ++k) {
+        float z = to_f(px.v[k]) * sc + sh;
+        if (RES) z += to_f(pr.v[k]);
+        if (RELU) z = fmaxf(z, 0.f);
+        py.v[k] = from_f<T>(z);
+      }
       *reinterpret_cast<Pack<T, V>*>(&y[e]) = py;
     }
   }
 }
 
-template <typename T, int V>
-__global__ void bn_elemt_nhwc(const T* __restrict__ x, T* __restrict__ y,
+template <typename T, int V, bool RELU, bool RES>
+__global__ void bn_elemt_nhwc(const T* __restrict__ x,
+                              const T* __restrict__ res, T* __restrict__ y,
                               const float* __restrict__ scale,
                               const float* __restrict__ shift, int64_t total,
                               int64_t C) {
@@ -339,32 +354,49 @@ __global__ void bn_elemt_nhwc(const T* __restrict__ x, T* __restrict__ y,
     const int64_t e = i * V;
     const int64_t c = e % C;
     if (V == 1) {
-      y[e] = from_f<T>(to_f(x[e]) * scale[c] + shift[c]);
+      float z = to_f(x[e]) * scale[c] + shift[c];
+      if (RES) z += to_f(res[e]);
+      if (RELU) z = fmaxf(z, 0.f);
+      y[e] = from_f<T>(z);
     } else {
       Pack<T, V> px = *reinterpret_cast<const Pack<T, V>*>(&x[e]);
       Pack<float, V> ps = *reinterpret_cast<const Pack<float, V>*>(&scale[c]);
       Pack<float, V> pb = *reinterpret_cast<const Pack<float, V>*>(&shift[c]);
+      Pack<T, V> pr;
+      if (RES) pr = *reinterpret_cast<const Pack<T, V>*>(&res[e]);
       Pack<T, V> py;
 #pragma unroll
-      for (int k = 0; k < V; ++k)
-        py.v[k] = from_f<T>(to_f(px.v[k]) * ps.v[k] + pb.v[k]);
+      for (int k = 0; k < V; ++k) {
+        float z = to_f(px.v[k]) * ps.v[k] + pb.v[k];
+        if (RES) z += to_f(pr.v[k]);
+        if (RELU) z = fmaxf(z, 0.f);
+        py.v[k] = from_f<T>(z);
+      }
       *reinterpret_cast<Pack<T, V>*>(&y[e]) = py;
     }
   }
 }
 
 // =====================================================================
-// backward reduce partial: {sum_dy, sum_dy*(x-mean)} per channel
+// backward reduce partial: {sum_g, sum_g*(x-mean)} per channel, where
+// g = dy masked by the fused-ReLU gate when MASK: the pre-activation
+// z = scale[c]*x + shift[c] (+ res) is recomputed in-kernel, so the fused
+// forward never has to store a mask or the pre-activation tensor.
 // =====================================================================
-template <typename T, int V>
+template <typename T, int V, bool MASK, bool RES>
 __global__ void bn_bwd_reduce_partial_nchw(const T* __restrict__ dy,
                                            const T* __restrict__ x,
+                                           const T* __restrict__ res,
                                            const float* __restrict__ mean,
+                                           const float* __restrict__ scale,
+                                           const float* __restrict__ shift,
                                            double* __restrict__ ws, int64_t N,
                                            int64_t C, int64_t S, int64_t chunkN,
                                            int64_t chunkS) {
   const int64_t c = blockIdx.x;
   const float m = mean[c];
+  const float sc = MASK ? scale[c] : 0.f;
+  const float sh = MASK ? shift[c] : 0.f;
   const int64_t n0 = blockIdx.y * chunkN;
   const int64_t n1 = i64min(n0 + chunkN, N);
   const int64_t s0 = blockIdx.z * chunkS;
@@ -374,20 +406,34 @@ __global__ void bn_bwd_reduce_partial_nchw(const T* __restrict__ dy,
     const int64_t base = (n * C + c) * S;
     if (V == 1) {
       for (int64_t s = s0 + threadIdx.x; s < s1; s += blockDim.x) {
-        const float g = to_f(dy[base + s]);
+        float g = to_f(dy[base + s]);
+        const float xv = to_f(x[base + s]);
+        if (MASK) {
+          float z = sc * xv + sh;
+          if (RES) z += to_f(res[base + s]);
+          if (z <= 0.f) g = 0.f;
+        }
         a += g;
-        b += (double)g * (to_f(x[base + s]) - m);
+        b += (double)g * (xv - m);
       }
     } else {
       for (int64_t s = s0 + (int64_t)threadIdx.x * V; s < s1;
            s += (int64_t)blockDim.x * V) {
         Pack<T, V> pg = *reinterpret_cast<const Pack<T, V>*>(&dy[base + s]);
         Pack<T, V> px = *reinterpret_cast<const Pack<T, V>*>(&x[base + s]);
+        Pack<T, V> pr;
+        if (RES) pr = *reinterpret_cast<const Pack<T, V>*>(&res[base + s]);
 #pragma unroll
         for (int k = 0; k < V; ++k) {
-          const float g = to_f(pg.v[k]);
+          float g = to_f(pg.v[k]);
+          const float xv = to_f(px.v[k]);
+          if (MASK) {
+            float z = sc * xv + sh;
+            if (RES) z += to_f(pr.v[k]);
+            if (z <= 0.f) g = 0.f;
+          }
           a += g;
-          b += (double)g * (to_f(px.v[k]) - m);
+          b += (double)g * (xv - m);
         }
       }
     }
@@ -402,13 +448,17 @@ __global__ void bn_bwd_reduce_partial_nchw(const T* __restrict__ dy,
   }
 }
 
-template <typename T>
+template <typename T, bool MASK, bool RES>
 __global__ void bn_bwd_reduce_partial_nchw_flat(
     const T* __restrict__ dy, const T* __restrict__ x,
-    const float* __restrict__ mean, double* __restrict__ ws, int64_t N,
-    int64_t C, int64_t S, int64_t chunk_len) {
+    const T* __restrict__ res, const float* __restrict__ mean,
+    const float* __restrict__ scale, const float* __restrict__ shift,
+    double* __restrict__ ws, int64_t N, int64_t C, int64_t S,
+    int64_t chunk_len) {
   const int64_t c = blockIdx.x;
   const float m = mean[c];
+  const float sc = MASK ? scale[c] : 0.f;
+  const float sh = MASK ? shift[c] : 0.f;
   const int64_t NS = N * S;
   const int64_t p0 = (int64_t)blockIdx.y * chunk_len;
   const int64_t p1 = i64min(p0 + chunk_len, NS);
@@ -416,9 +466,15 @@ __global__ void bn_bwd_reduce_partial_nchw_flat(
   for (int64_t p = p0 + threadIdx.x; p < p1; p += blockDim.x) {
     const int64_t n = p / S, s = p - n * S;
     const int64_t e = (n * C + c) * S + s;
-    const float g = to_f(dy[e]);
+    float g = to_f(dy[e]);
+    const float xv = to_f(x[e]);
+    if (MASK) {
+      float z = sc * xv + sh;
+      if (RES) z += to_f(res[e]);
+      if (z <= 0.f) g = 0.f;
+    }
     a += g;
-    b += (double)g * (to_f(x[e]) - m);
+    b += (double)g * (xv - m);
   }
   __shared__ double lds[2 * (MSBN_BLOCK / MSBN_WAVE)];
   block_reduce_pair(a, b, lds);
@@ -429,10 +485,13 @@ __global__ void bn_bwd_reduce_partial_nchw_flat(
   }
 }
 
-template <typename T, int V>
+template <typename T, int V, bool MASK, bool RES>
 __global__ void bn_bwd_reduce_partial_nhwc(const T* __restrict__ dy,
                                            const T* __restrict__ x,
+                                           const T* __restrict__ res,
                                            const float* __restrict__ mean,
+                                           const float* __restrict__ scale,
+                                           const float* __restrict__ shift,
                                            double* __restrict__ ws,
                                            int64_t rows, int64_t C,
                                            int64_t chunk_rows, int lpr) {
@@ -444,25 +503,39 @@ __global__ void bn_bwd_reduce_partial_nhwc(const T* __restrict__ dy,
   const bool inb = active && (c + V <= C);
 
   double a[V], b[V];
-  float m[V];
+  float m[V], scv[V], shv[V];
 #pragma unroll
   for (int k = 0; k < V; ++k) {
     a[k] = b[k] = 0.0;
-    m[k] = 0.f;
+    m[k] = scv[k] = shv[k] = 0.f;
   }
   if (inb) {
 #pragma unroll
-    for (int k = 0; k < V; ++k) m[k] = mean[c + k];
+    for (int k = 0; k < V; ++k) {
+      m[k] = mean[c + k];
+      if (MASK) {
+        scv[k] = scale[c + k];
+        shv[k] = shift[c + k];
+      }
+    }
     const int64_t r0 = (int64_t)blockIdx.y * chunk_rows;
     const int64_t r1 = i64min(r0 + chunk_rows, rows);
     for (int64_t r = r0 + rowoff; r < r1; r += rpi) {
       Pack<T, V> pg = *reinterpret_cast<const Pack<T, V>*>(&dy[r * C + c]);
       Pack<T, V> px = *reinterpret_cast<const Pack<T, V>*>(&x[r * C + c]);
+      Pack<T, V> pr;
+      if (RES) pr = *reinterpret_cast<const Pack<T, V>*>(&res[r * C + c]);
 #pragma unroll
       for (int k = 0; k < V; ++k) {
-        const float g = to_f(pg.v[k]);
+        float g = to_f(pg.v[k]);
+        const float xv = to_f(px.v[k]);
+        if (MASK) {
+          float z = scv[k] * xv + shv[k];
+          if (RES) z += to_f(pr.v[k]);
+          if (z <= 0.f) g = 0.f;
+        }
         a[k] += g;
-        b[k] += (double)g * (to_f(px.v[k]) - m[k]);
+        b[k] += (double)g * (xv - m[k]);
       }
     }
   }
@@ -524,61 +597,117 @@ __global__ void bn_bwd_reduce_finalize(const double* __restrict__ ws,
 }
 
 // =====================================================================
-// backward elementwise: dx = a[c]*dy + b[c]*x + d[c]
+// backward elementwise: dx = a[c]*g + b[c]*x + d[c], with the fused-ReLU
+// gate recomputed in-kernel when MASK (g = z>0 ? dy : 0); RESG additionally
+// writes the residual-branch gradient dres = g (the add node's pass-through).
 // =====================================================================
-template <typename T, int V>
+template <typename T, int V, bool MASK, bool RES, bool RESG>
 __global__ void bn_bwd_elemt_nchw(const T* __restrict__ dy,
-                                  const T* __restrict__ x, T* __restrict__ dx,
+                                  const T* __restrict__ x,
+                                  const T* __restrict__ res,
+                                  T* __restrict__ dx, T* __restrict__ dres,
                                   const float* __restrict__ ca,
                                   const float* __restrict__ cb,
-                                  const float* __restrict__ cd, int64_t total,
-                                  int64_t C, int64_t S) {
+                                  const float* __restrict__ cd,
+                                  const float* __restrict__ scale,
+                                  const float* __restrict__ shift,
+                                  int64_t total, int64_t C, int64_t S) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
        i * V < total; i += stride) {
     const int64_t e = i * V;
     const int64_t c = (e / S) % C;
     const float A = ca[c], B = cb[c], D = cd[c];
+    const float sc = MASK ? scale[c] : 0.f;
+    const float sh = MASK ? shift[c] : 0.f;
     if (V == 1) {
-      dx[e] = from_f<T>(A * to_f(dy[e]) + B * to_f(x[e]) + D);
+      float g = to_f(dy[e]);
+      const float xv = to_f(x[e]);
+      if (MASK) {
+        float z = sc * xv + sh;
+        if (RES) z += to_f(res[e]);
+        if (z <= 0.f) g = 0.f;
+      }
+      dx[e] = from_f<T>(A * g + B * xv + D);
+      if (RESG) dres[e] = from_f<T>(g);
     } else {
       Pack<T, V> pg = *reinterpret_cast<const Pack<T, V>*>(&dy[e]);
       Pack<T, V> px = *reinterpret_cast<const Pack<T, V>*>(&x[e]);
-      Pack<T, V> po;
+      Pack<T, V> pr;
+      if (RES) pr = *reinterpret_cast<const Pack<T, V>*>(&res[e]);
+      Pack<T, V> po, pq;
 #pragma unroll
-      for (int k = 0; k < V; ++k)
-        po.v[k] = from_f<T>(A * to_f(pg.v[k]) + B * to_f(px.v[k]) + D);
+      for (int k = 0; k < V; ++k) {
+        float g = to_f(pg.v[k]);
+        const float xv = to_f(px.v[k]);
+        if (MASK) {
+          float z = sc * xv + sh;
+          if (RES) z += to_f(pr.v[k]);
+          if (z <= 0.f) g = 0.f;
+        }
+        po.v[k] = from_f<T>(A * g + B * xv + D);
+        if (RESG) pq.v[k] = from_f<T>(g);
+      }
       *reinterpret_cast<Pack<T, V>*>(&dx[e]) = po;
+      if (RESG) *reinterpret_cast<Pack<T, V>*>(&dres[e]) = pq;
     }
   }
 }
 
-template <typename T, int V>
+template <typename T, int V, bool MASK, bool RES, bool RESG>
 __global__ void bn_bwd_elemt_nhwc(const T* __restrict__ dy,
-                                  const T* __restrict__ x, T* __restrict__ dx,
+                                  const T* __restrict__ x,
+                                  const T* __restrict__ res,
+                                  T* __restrict__ dx, T* __restrict__ dres,
                                   const float* __restrict__ ca,
                                   const float* __restrict__ cb,
-                                  const float* __restrict__ cd, int64_t total,
-                                  int64_t C) {
+                                  const float* __restrict__ cd,
+                                  const float* __restrict__ scale,
+                                  const float* __restrict__ shift,
+                                  int64_t total, int64_t C) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
        i * V < total; i += stride) {
     const int64_t e = i * V;
     const int64_t c = e % C;
     if (V == 1) {
-      dx[e] = from_f<T>(ca[c] * to_f(dy[e]) + cb[c] * to_f(x[e]) + cd[c]);
+      float g = to_f(dy[e]);
+      const float xv = to_f(x[e]);
+      if (MASK) {
+        float z = scale[c] * xv + shift[c];
+        if (RES) z += to_f(res[e]);
+        if (z <= 0.f) g = 0.f;
+      }
+      dx[e] = from_f<T>(ca[c] * g + cb[c] * xv + cd[c]);
+      if (RESG) dres[e] = from_f<T>(g);
     } else {
       Pack<T, V> pg = *reinterpret_cast<const Pack<T, V>*>(&dy[e]);
       Pack<T, V> px = *reinterpret_cast<const Pack<T, V>*>(&x[e]);
       Pack<float, V> pa = *reinterpret_cast<const Pack<float, V>*>(&ca[c]);
       Pack<float, V> pb = *reinterpret_cast<const Pack<float, V>*>(&cb[c]);
       Pack<float, V> pd = *reinterpret_cast<const Pack<float, V>*>(&cd[c]);
-      Pack<T, V> po;
+      Pack<float, V> psc, psh;
+      if (MASK) {
+        psc = *reinterpret_cast<const Pack<float, V>*>(&scale[c]);
+        psh = *reinterpret_cast<const Pack<float, V>*>(&shift[c]);
+      }
+      Pack<T, V> pr;
+      if (RES) pr = *reinterpret_cast<const Pack<T, V>*>(&res[e]);
+      Pack<T, V> po, pq;
 #pragma unroll
-      for (int k = 0; k < V; ++k)
-        po.v[k] =
-            from_f<T>(pa.v[k] * to_f(pg.v[k]) + pb.v[k] * to_f(px.v[k]) + pd.v[k]);
+      for (int k = 0; k < V; ++k) {
+        float g = to_f(pg.v[k]);
+        const float xv = to_f(px.v[k]);
+        if (MASK) {
+          float z = psc.v[k] * xv + psh.v[k];
+          if (RES) z += to_f(pr.v[k]);
+          if (z <= 0.f) g = 0.f;
+        }
+        po.v[k] = from_f<T>(pa.v[k] * g + pb.v[k] * xv + pd.v[k]);
+        if (RESG) pq.v[k] = from_f<T>(g);
+      }
       *reinterpret_cast<Pack<T, V>*>(&dx[e]) = po;
+      if (RESG) *reinterpret_cast<Pack<T, V>*>(&dres[e]) = pq;
     }
   }
 }
@@ -944,25 +1073,20 @@ std::tuple<at::Tensor, at::Tensor> batch_norm_gather_stats_with_counts(
   return {std::get<0>(out), std::get<1>(out)};
 }
 
-at::Tensor batch_norm_elemt(const at::Tensor& input,
-                            const c10::optional<at::Tensor>& weight,
-                            const c10::optional<at::Tensor>& bias,
-                            const at::Tensor& mean, const at::Tensor& invstd,
-                            double eps) {
-  (void)eps;  // invstd already folds eps
-  const Layout L = get_layout(input);
-  auto out = at::empty_like(input);
-  if (input.numel() == 0) return out;
-  auto stream = cur_stream();
-  auto f32 = input.options().dtype(at::kFloat);
-  auto coefs = at::empty({2 * L.C}, f32);
+namespace {
+// compute per-channel (scale, shift) into one [2C] fp32 buffer
+at::Tensor make_fwd_coefs(const at::Tensor& mean, const at::Tensor& invstd,
+                          const c10::optional<at::Tensor>& weight,
+                          const c10::optional<at::Tensor>& bias, int64_t C) {
+  auto coefs = at::empty({2 * C}, mean.options().dtype(at::kFloat));
   float* scale = coefs.data_ptr<float>();
-  float* shift = scale + L.C;
-  const int cgrid = (int)cdiv(L.C, MSBN_BLOCK);
+  float* shift = scale + C;
+  auto stream = cur_stream();
+  const int cgrid = (int)cdiv(C, MSBN_BLOCK);
   const auto wtype = weight.has_value() ? weight->scalar_type()
                      : bias.has_value() ? bias->scalar_type()
                                         : at::kFloat;
-  MSBN_DISPATCH_RSTAT(wtype, "bn_elemt", [&] {
+  MSBN_DISPATCH_RSTAT(wtype, "bn_affine", [&] {
     const rstat_t* w =
         weight.has_value()
             ? reinterpret_cast<const rstat_t*>(weight->data_ptr())
@@ -972,41 +1096,91 @@ at::Tensor batch_norm_elemt(const at::Tensor& input,
                            : nullptr;
     hipLaunchKernelGGL((bn_affine_fwd<rstat_t>), dim3(cgrid), dim3(MSBN_BLOCK),
                        0, stream, mean.data_ptr<float>(),
-                       invstd.data_ptr<float>(), w, b, L.C, scale, shift);
+                       invstd.data_ptr<float>(), w, b, C, scale, shift);
   });
+  return coefs;
+}
+
+#define MSBN_DISPATCH_BOOL(FLAG, NAME, ...)                                  \
+  [&] {                                                                      \
+    if (FLAG) {                                                              \
+      constexpr bool NAME = true;                                            \
+      return __VA_ARGS__();                                                  \
+    } else {                                                                 \
+      constexpr bool NAME = false;                                           \
+      return __VA_ARGS__();                                                  \
+    }                                                                        \
+  }()
+}  // namespace
+
+at::Tensor batch_norm_elemt_act(const at::Tensor& input,
+                                const c10::optional<at::Tensor>& residual,
+                                const c10::optional<at::Tensor>& weight,
+                                const c10::optional<at::Tensor>& bias,
+                                const at::Tensor& mean,
+                                const at::Tensor& invstd, bool relu) {
+  const Layout L = get_layout(input);
+  auto out = at::empty_like(input);
+  if (input.numel() == 0) return out;
+  auto stream = cur_stream();
+  auto coefs = make_fwd_coefs(mean, invstd, weight, bias, L.C);
+  float* scale = coefs.data_ptr<float>();
+  float* shift = scale + L.C;
+  const bool has_res = residual.has_value();
+  if (has_res) {
+    TORCH_CHECK(residual->sizes() == input.sizes() &&
+                    residual->scalar_type() == input.scalar_type(),
+                "residual must match input shape/dtype");
+  }
 
   const int64_t total = input.numel();
   MSBN_DISPATCH_FLOAT_TYPES(input.scalar_type(), "bn_elemt", [&] {
     const native_t* x =
         reinterpret_cast<const native_t*>(input.data_ptr<scalar_t>());
+    const native_t* res =
+        has_res ? reinterpret_cast<const native_t*>(residual->data_ptr())
+                : nullptr;
     native_t* y = reinterpret_cast<native_t*>(out.data_ptr<scalar_t>());
     constexpr int VMAX = 16 / (int)sizeof(native_t);
-    if (!L.nhwc) {
-      const int v = pick_v<native_t>(x, y, nullptr, L.S);
-      const int grid = elemt_grid(total, v);
-      MSBN_DISPATCH_V(v, VMAX, [&] {
-        hipLaunchKernelGGL((bn_elemt_nchw<native_t, VV>), dim3(grid),
-                           dim3(MSBN_BLOCK), 0, stream, x, y, scale, shift,
-                           total, L.C, L.S);
+    const int64_t inner = L.nhwc ? L.C : L.S;
+    const int v = pick_v<native_t>(x, y, res, inner);
+    const int grid = elemt_grid(total, v);
+    MSBN_DISPATCH_V(v, VMAX, [&] {
+      MSBN_DISPATCH_BOOL(relu, RELU_, [&] {
+        MSBN_DISPATCH_BOOL(has_res, RES_, [&] {
+          if (!L.nhwc) {
+            hipLaunchKernelGGL((bn_elemt_nchw<native_t, VV, RELU_, RES_>),
+                               dim3(grid), dim3(MSBN_BLOCK), 0, stream, x, res,
+                               y, scale, shift, total, L.C, L.S);
+          } else {
+            hipLaunchKernelGGL((bn_elemt_nhwc<native_t, VV, RELU_, RES_>),
+                               dim3(grid), dim3(MSBN_BLOCK), 0, stream, x, res,
+                               y, scale, shift, total, L.C);
+          }
+        });
       });
-    } else {
-      const int v = pick_v<native_t>(x, y, nullptr, L.C);
-      const int grid = elemt_grid(total, v);
-      MSBN_DISPATCH_V(v, VMAX, [&] {
-        hipLaunchKernelGGL((bn_elemt_nhwc<native_t, VV>), dim3(grid),
-                           dim3(MSBN_BLOCK), 0, stream, x, y, scale, shift,
-                           total, L.C);
-      });
-    }
+    });
   });
   return out;
 }
 
+at::Tensor batch_norm_elemt(const at::Tensor& input,
+                            const c10::optional<at::Tensor>& weight,
+                            const c10::optional<at::Tensor>& bias,
+                            const at::Tensor& mean, const at::Tensor& invstd,
+                            double eps) {
+  (void)eps;  // invstd already folds eps
+  return batch_norm_elemt_act(input, c10::nullopt, weight, bias, mean, invstd,
+                              false);
+}
+
 std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor>
-batch_norm_backward_reduce(const at::Tensor& grad_out, const at::Tensor& input,
-                           const at::Tensor& mean, const at::Tensor& invstd,
-                           const c10::optional<at::Tensor>& weight, bool input_g,
-                           bool weight_g, bool bias_g) {
+batch_norm_backward_reduce_act(
+    const at::Tensor& grad_out, const at::Tensor& input,
+    const c10::optional<at::Tensor>& residual, const at::Tensor& mean,
+    const at::Tensor& invstd, const c10::optional<at::Tensor>& weight,
+    const c10::optional<at::Tensor>& bias, bool relu_mask, bool input_g,
+    bool weight_g, bool bias_g) {
   const Layout L = get_layout(input);
   TORCH_CHECK(grad_out.sizes() == input.sizes(), "grad_out/input shape mismatch");
   auto f32 = input.options().dtype(at::kFloat);
@@ -1021,51 +1195,71 @@ batch_norm_backward_reduce(const at::Tensor& grad_out, const at::Tensor& input,
   if (weight_g) grad_weight = at::empty({L.C}, wopts);
   if (bias_g) grad_bias = at::empty({L.C}, wopts);
 
+  const bool has_res = residual.has_value();
+  float* scale = nullptr;
+  float* shift = nullptr;
+  at::Tensor coefs;
+  if (relu_mask) {
+    coefs = make_fwd_coefs(mean, invstd, weight, bias, L.C);
+    scale = coefs.data_ptr<float>();
+    shift = scale + L.C;
+  }
+
   auto stream = cur_stream();
   MSBN_DISPATCH_FLOAT_TYPES(input.scalar_type(), "bn_bwd_reduce", [&] {
     const native_t* x =
         reinterpret_cast<const native_t*>(input.data_ptr<scalar_t>());
     const native_t* dy =
         reinterpret_cast<const native_t*>(grad_out.data_ptr<scalar_t>());
+    const native_t* res =
+        has_res ? reinterpret_cast<const native_t*>(residual->data_ptr())
+                : nullptr;
     constexpr int VMAX = 16 / (int)sizeof(native_t);
     at::Tensor ws;
     int nchunks = 0;
-    if (!L.nhwc) {
-      const int v = pick_v<native_t>(x, dy, nullptr, L.S);
-      if (v == 1) {
-        auto g = flat_grid(L.C, L.rows, kTargetBlocks);
-        nchunks = g.nchunks;
-        ws = at::empty({(int64_t)nchunks * L.C * 2},
-                       input.options().dtype(at::kDouble));
-        hipLaunchKernelGGL((bn_bwd_reduce_partial_nchw_flat<native_t>), g.grid,
-                           dim3(MSBN_BLOCK), 0, stream, dy, x,
-                           mean.data_ptr<float>(), ws.data_ptr<double>(), L.N,
-                           L.C, L.S, g.chunk_len);
-      } else {
-      auto g = nchw_grid(L.N, L.C, L.S, v);
-      nchunks = g.nchunks;
-      ws = at::empty({(int64_t)nchunks * L.C * 2},
-                     input.options().dtype(at::kDouble));
-      MSBN_DISPATCH_V(v, VMAX, [&] {
-        hipLaunchKernelGGL((bn_bwd_reduce_partial_nchw<native_t, VV>), g.grid,
-                           dim3(MSBN_BLOCK), 0, stream, dy, x,
-                           mean.data_ptr<float>(), ws.data_ptr<double>(), L.N,
-                           L.C, L.S, g.chunkN, g.chunkS);
+    MSBN_DISPATCH_BOOL(relu_mask, MASK_, [&] {
+      MSBN_DISPATCH_BOOL(has_res, RES_, [&] {
+        if (!L.nhwc) {
+          const int v = pick_v<native_t>(x, dy, res, L.S);
+          if (v == 1) {
+            auto g = flat_grid(L.C, L.rows, kTargetBlocks);
+            nchunks = g.nchunks;
+            ws = at::empty({(int64_t)nchunks * L.C * 2},
+                           input.options().dtype(at::kDouble));
+            hipLaunchKernelGGL(
+                (bn_bwd_reduce_partial_nchw_flat<native_t, MASK_, RES_>),
+                g.grid, dim3(MSBN_BLOCK), 0, stream, dy, x, res,
+                mean.data_ptr<float>(), scale, shift, ws.data_ptr<double>(),
+                L.N, L.C, L.S, g.chunk_len);
+          } else {
+            auto g = nchw_grid(L.N, L.C, L.S, v);
+            nchunks = g.nchunks;
+            ws = at::empty({(int64_t)nchunks * L.C * 2},
+                           input.options().dtype(at::kDouble));
+            MSBN_DISPATCH_V(v, VMAX, [&] {
+              hipLaunchKernelGGL(
+                  (bn_bwd_reduce_partial_nchw<native_t, VV, MASK_, RES_>),
+                  g.grid, dim3(MSBN_BLOCK), 0, stream, dy, x, res,
+                  mean.data_ptr<float>(), scale, shift, ws.data_ptr<double>(),
+                  L.N, L.C, L.S, g.chunkN, g.chunkS);
+            });
+          }
+        } else {
+          const int v = pick_v<native_t>(x, dy, res, L.C);
+          auto g = nhwc_grid(L.rows, L.C, v);
+          nchunks = g.nchunks;
+          ws = at::empty({(int64_t)nchunks * L.C * 2},
+                         input.options().dtype(at::kDouble));
+          MSBN_DISPATCH_V(v, VMAX, [&] {
+            hipLaunchKernelGGL(
+                (bn_bwd_reduce_partial_nhwc<native_t, VV, MASK_, RES_>),
+                g.grid, dim3(MSBN_BLOCK), 0, stream, dy, x, res,
+                mean.data_ptr<float>(), scale, shift, ws.data_ptr<double>(),
+                L.rows, L.C, g.chunk_rows, g.lpr);
+          });
+        }
       });
-      }
-    } else {
-      const int v = pick_v<native_t>(x, dy, nullptr, L.C);
-      auto g = nhwc_grid(L.rows, L.C, v);
-      nchunks = g.nchunks;
-      ws = at::empty({(int64_t)nchunks * L.C * 2},
-                     input.options().dtype(at::kDouble));
-      MSBN_DISPATCH_V(v, VMAX, [&] {
-        hipLaunchKernelGGL((bn_bwd_reduce_partial_nhwc<native_t, VV>), g.grid,
-                           dim3(MSBN_BLOCK), 0, stream, dy, x,
-                           mean.data_ptr<float>(), ws.data_ptr<double>(),
-                           L.rows, L.C, g.chunk_rows, g.lpr);
-      });
-    }
+    });
     const int fgrid = (int)cdiv(L.C, kFinalizeWavesPerBlock);
     MSBN_DISPATCH_RSTAT(wtype, "bn_bwd_reduce", [&] {
       rstat_t* gw = weight_g ? reinterpret_cast<rstat_t*>(grad_weight.data_ptr())
@@ -1083,14 +1277,28 @@ batch_norm_backward_reduce(const at::Tensor& grad_out, const at::Tensor& input,
   return {sum_dy, sum_dy_xmu, grad_weight, grad_bias};
 }
 
-at::Tensor batch_norm_backward_elemt(
-    const at::Tensor& grad_out, const at::Tensor& input, const at::Tensor& mean,
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor>
+batch_norm_backward_reduce(const at::Tensor& grad_out, const at::Tensor& input,
+                           const at::Tensor& mean, const at::Tensor& invstd,
+                           const c10::optional<at::Tensor>& weight, bool input_g,
+                           bool weight_g, bool bias_g) {
+  return batch_norm_backward_reduce_act(grad_out, input, c10::nullopt, mean,
+                                        invstd, weight, c10::nullopt, false,
+                                        input_g, weight_g, bias_g);
+}
+
+std::tuple<at::Tensor, at::Tensor> batch_norm_backward_elemt_act(
+    const at::Tensor& grad_out, const at::Tensor& input,
+    const c10::optional<at::Tensor>& residual, const at::Tensor& mean,
     const at::Tensor& invstd, const c10::optional<at::Tensor>& weight,
-    const at::Tensor& sum_dy, const at::Tensor& sum_dy_xmu,
-    const at::Tensor& count) {
+    const c10::optional<at::Tensor>& bias, const at::Tensor& sum_dy,
+    const at::Tensor& sum_dy_xmu, const at::Tensor& count, bool relu_mask,
+    bool want_res_grad) {
   const Layout L = get_layout(input);
   auto dx = at::empty_like(grad_out);
-  if (input.numel() == 0) return dx;
+  at::Tensor dres;
+  if (want_res_grad) dres = at::empty_like(grad_out);
+  if (input.numel() == 0) return {dx, dres};
   // total count as a device fp32 scalar (no host sync)
   at::Tensor count_sum;
   if (count.numel() == 1 && count.scalar_type() == at::kFloat) {
@@ -1118,6 +1326,15 @@ at::Tensor batch_norm_backward_elemt(
                        sum_dy_xmu.data_ptr<float>(),
                        count_sum.data_ptr<float>(), L.C, ca, cb, cd);
   });
+  const bool has_res = residual.has_value();
+  float* scale = nullptr;
+  float* shift = nullptr;
+  at::Tensor fcoefs;
+  if (relu_mask) {
+    fcoefs = make_fwd_coefs(mean, invstd, weight, bias, L.C);
+    scale = fcoefs.data_ptr<float>();
+    shift = scale + L.C;
+  }
 
   const int64_t total = input.numel();
   MSBN_DISPATCH_FLOAT_TYPES(input.scalar_type(), "bn_bwd_elemt", [&] {
@@ -1125,27 +1342,48 @@ at::Tensor batch_norm_backward_elemt(
         reinterpret_cast<const native_t*>(input.data_ptr<scalar_t>());
     const native_t* dy =
         reinterpret_cast<const native_t*>(grad_out.data_ptr<scalar_t>());
+    const native_t* res =
+        has_res ? reinterpret_cast<const native_t*>(residual->data_ptr())
+                : nullptr;
     native_t* o = reinterpret_cast<native_t*>(dx.data_ptr<scalar_t>());
+    native_t* dr = want_res_grad
+                       ? reinterpret_cast<native_t*>(dres.data_ptr<scalar_t>())
+                       : nullptr;
     constexpr int VMAX = 16 / (int)sizeof(native_t);
-    if (!L.nhwc) {
-      const int v = pick_v<native_t>(x, dy, o, L.S);
-      const int grid = elemt_grid(total, v);
-      MSBN_DISPATCH_V(v, VMAX, [&] {
-        hipLaunchKernelGGL((bn_bwd_elemt_nchw<native_t, VV>), dim3(grid),
-                           dim3(MSBN_BLOCK), 0, stream, dy, x, o, ca, cb, cd,
-                           total, L.C, L.S);
+    const int64_t inner = L.nhwc ? L.C : L.S;
+    const int v = pick_v<native_t>(x, dy, o, inner);
+    const int grid = elemt_grid(total, v);
+    MSBN_DISPATCH_V(v, VMAX, [&] {
+      MSBN_DISPATCH_BOOL(relu_mask, MASK_, [&] {
+        MSBN_DISPATCH_BOOL(has_res, RES_, [&] {
+          MSBN_DISPATCH_BOOL(want_res_grad, RESG_, [&] {
+            if (!L.nhwc) {
+              hipLaunchKernelGGL(
+                  (bn_bwd_elemt_nchw<native_t, VV, MASK_, RES_, RESG_>),
+                  dim3(grid), dim3(MSBN_BLOCK), 0, stream, dy, x, res, o, dr,
+                  ca, cb, cd, scale, shift, total, L.C, L.S);
+            } else {
+              hipLaunchKernelGGL(
+                  (bn_bwd_elemt_nhwc<native_t, VV, MASK_, RES_, RESG_>),
+                  dim3(grid), dim3(MSBN_BLOCK), 0, stream, dy, x, res, o, dr,
+                  ca, cb, cd, scale, shift, total, L.C);
+            }
+          });
+        });
       });
-    } else {
-      const int v = pick_v<native_t>(x, dy, o, L.C);
-      const int grid = elemt_grid(total, v);
-      MSBN_DISPATCH_V(v, VMAX, [&] {
-        hipLaunchKernelGGL((bn_bwd_elemt_nhwc<native_t, VV>), dim3(grid),
-                           dim3(MSBN_BLOCK), 0, stream, dy, x, o, ca, cb, cd,
-                           total, L.C);
-      });
-    }
+    });
   });
-  return dx;
+  return {dx, dres};
+}
+
+at::Tensor batch_norm_backward_elemt(
+    const at::Tensor& grad_out, const at::Tensor& input, const at::Tensor& mean,
+    const at::Tensor& invstd, const c10::optional<at::Tensor>& weight,
+    const at::Tensor& sum_dy, const at::Tensor& sum_dy_xmu,
+    const at::Tensor& count) {
+  return std::get<0>(batch_norm_backward_elemt_act(
+      grad_out, input, c10::nullopt, mean, invstd, weight, c10::nullopt,
+      sum_dy, sum_dy_xmu, count, false, false));
 }
 
 }  // namespace msbn

@@ -55,4 +55,33 @@ at::Tensor batch_norm_backward_elemt(
     const at::Tensor& sum_dy, const at::Tensor& sum_dy_xmu,
     const at::Tensor& count);
 
+// ---- fused BN(+residual add)(+ReLU) epilogue path -------------------------
+// Forward: y = relu?(x*scale + shift [+ residual]).  Backward recomputes the
+// ReLU gate in-kernel from (x, scale, shift[, residual]) — no mask tensor and
+// no separate clamp/add/threshold_backward kernels or their full-tensor
+// round trips.
+at::Tensor batch_norm_elemt_act(const at::Tensor& input,
+                                const c10::optional<at::Tensor>& residual,
+                                const c10::optional<at::Tensor>& weight,
+                                const c10::optional<at::Tensor>& bias,
+                                const at::Tensor& mean,
+                                const at::Tensor& invstd, bool relu);
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor>
+batch_norm_backward_reduce_act(
+    const at::Tensor& grad_out, const at::Tensor& input,
+    const c10::optional<at::Tensor>& residual, const at::Tensor& mean,
+    const at::Tensor& invstd, const c10::optional<at::Tensor>& weight,
+    const c10::optional<at::Tensor>& bias, bool relu_mask, bool input_g,
+    bool weight_g, bool bias_g);
+
+// returns (grad_input, grad_residual-or-undefined)
+std::tuple<at::Tensor, at::Tensor> batch_norm_backward_elemt_act(
+    const at::Tensor& grad_out, const at::Tensor& input,
+    const c10::optional<at::Tensor>& residual, const at::Tensor& mean,
+    const at::Tensor& invstd, const c10::optional<at::Tensor>& weight,
+    const c10::optional<at::Tensor>& bias, const at::Tensor& sum_dy,
+    const at::Tensor& sum_dy_xmu, const at::Tensor& count, bool relu_mask,
+    bool want_res_grad);
+
 }  // namespace msbn
