@@ -46,6 +46,18 @@ def _contig(t: torch.Tensor) -> torch.Tensor:
     return t.contiguous()
 
 
+def _match_layout(t: torch.Tensor, ref: torch.Tensor) -> torch.Tensor:
+    """Coerce t to ref's memory format (the elementwise/reduce kernels index
+    both tensors with ONE layout)."""
+    if ref.dim() == 4 and ref.is_contiguous(memory_format=torch.channels_last):
+        return t.contiguous(memory_format=torch.channels_last)
+    if ref.dim() == 5 and ref.is_contiguous(
+        memory_format=torch.channels_last_3d
+    ):
+        return t.contiguous(memory_format=torch.channels_last_3d)
+    return t.contiguous()
+
+
 def compute_sync_stats(
     input: torch.Tensor,
     eps: float,
@@ -139,12 +151,8 @@ class SyncBatchNormFunction(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, grad_output: torch.Tensor):
-        if not (
-            grad_output.is_contiguous(memory_format=torch.channels_last)
-            or grad_output.is_contiguous(memory_format=torch.channels_last_3d)
-        ):
-            grad_output = grad_output.contiguous()
         input, weight, mean, invstd, count_sum = ctx.saved_tensors
+        grad_output = _match_layout(grad_output, input)
         process_group = ctx.process_group
         world_size = ctx.world_size
         need_input_g, need_weight_g, need_bias_g = ctx.needs_input_grad[0:3]

@@ -18,7 +18,7 @@ import torch.distributed as dist
 
 from msbn import ops
 from msbn.nn.batchnorm import SyncBatchNorm, _momentum_factor
-from msbn.nn.functions import _contig, compute_sync_stats
+from msbn.nn.functions import _contig, _match_layout, compute_sync_stats
 
 
 class SyncBatchNormActFunction(torch.autograd.Function):
@@ -39,7 +39,7 @@ class SyncBatchNormActFunction(torch.autograd.Function):
     ):
         input = _contig(input)
         if residual is not None:
-            residual = _contig(residual)
+            residual = _match_layout(residual, input)
         if weight is not None:
             weight = weight.contiguous()
         if bias is not None:
@@ -60,10 +60,10 @@ class SyncBatchNormActFunction(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, grad_output: torch.Tensor):
-        grad_output = _contig(grad_output)
         input, residual, weight, bias, mean, invstd, count_sum = (
             ctx.saved_tensors
         )
+        grad_output = _match_layout(grad_output, input)
         relu = ctx.relu
         process_group = ctx.process_group
         world_size = ctx.world_size

@@ -1129,8 +1129,9 @@ at::Tensor batch_norm_elemt_act(const at::Tensor& input,
   const bool has_res = residual.has_value();
   if (has_res) {
     TORCH_CHECK(residual->sizes() == input.sizes() &&
+                    residual->strides() == input.strides() &&
                     residual->scalar_type() == input.scalar_type(),
-                "residual must match input shape/dtype");
+                "residual must match input shape/strides/dtype");
   }
 
   const int64_t total = input.numel();
@@ -1182,7 +1183,13 @@ batch_norm_backward_reduce_act(
     const c10::optional<at::Tensor>& bias, bool relu_mask, bool input_g,
     bool weight_g, bool bias_g) {
   const Layout L = get_layout(input);
-  TORCH_CHECK(grad_out.sizes() == input.sizes(), "grad_out/input shape mismatch");
+  TORCH_CHECK(grad_out.sizes() == input.sizes() &&
+                  grad_out.strides() == input.strides(),
+              "grad_out must match input shape and layout");
+  TORCH_CHECK(!residual.has_value() ||
+                  (residual->sizes() == input.sizes() &&
+                   residual->strides() == input.strides()),
+              "residual must match input shape and layout");
   auto f32 = input.options().dtype(at::kFloat);
   // ONE contiguous buffer for [sum_dy | sum_dy_xmu] -> single all_reduce.
   auto combined = at::empty({2 * L.C}, f32);

@@ -75,9 +75,9 @@ def test_fused_function_gpu_vs_cpu(dtype, channels_last):
                 b.grad.float().cpu(), rm.cpu(), rv.cpu())
 
     got = run(DEV, dtype, channels_last)
-    want = run("cpu", torch.float32, False)
+    want = run("cpu", dtype, False)
     tol = dict(atol=1e-4, rtol=1e-4) if dtype == torch.float32 else \
-        dict(atol=5e-2, rtol=5e-2)
+        dict(atol=2e-2, rtol=2e-2)
     for g, w_, name in zip(got, want,
                            ["y", "dx", "dres", "dw", "db", "rm", "rv"]):
         torch.testing.assert_close(g, w_, **tol), name
