@@ -1,0 +1,118 @@
+#!/usr/bin/env python3
+"""DCGAN G+D with SyncBN (BASELINE.json config 4: small-per-GPU-batch GAN).
+
+    python benchmarks/bench_dcgan.py [--steps 20 --warmup 5 --batch-size 64]
+    torchrun --nproc-per-node 4 benchmarks/bench_dcgan.py ...
+
+Rank 0 prints one JSON line (images/sec whole job, G+D step)."""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+import torch.distributed as dist
+
+import msbn
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--batch-size", type=int, default=64)
+    p.add_argument("--image-size", type=int, default=64)
+    p.add_argument("--local_rank", "--local-rank", type=int,
+                   default=int(os.environ.get("LOCAL_RANK", 0)),
+                   dest="local_rank")
+    args = p.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    use_cuda = torch.cuda.is_available()
+    device = torch.device(f"cuda:{args.local_rank}" if use_cuda else "cpu")
+    if use_cuda:
+        torch.cuda.set_device(device)
+    if world > 1:
+        dist.init_process_group("nccl" if use_cuda else "gloo",
+                                init_method="env://", world_size=world,
+                                rank=rank)
+
+    torch.manual_seed(7)
+    G = msbn.convert_sync_batchnorm(msbn.models.Generator()).to(device)
+    D = msbn.convert_sync_batchnorm(msbn.models.Discriminator()).to(device)
+    if world > 1:
+        G = msbn.parallel.DistributedDataParallel(
+            G, device_ids=[args.local_rank] if use_cuda else None)
+        D = msbn.parallel.DistributedDataParallel(
+            D, device_ids=[args.local_rank] if use_cuda else None)
+    optG = torch.optim.Adam(G.parameters(), lr=2e-4, betas=(0.5, 0.999))
+    optD = torch.optim.Adam(D.parameters(), lr=2e-4, betas=(0.5, 0.999))
+    bce = torch.nn.BCEWithLogitsLoss()
+
+    bs = args.batch_size
+    real = torch.randn(bs, 3, args.image_size, args.image_size, device=device)
+    ones = torch.ones(bs, 1, device=device)
+    zeros = torch.zeros(bs, 1, device=device)
+
+    def step():
+        z = torch.randn(bs, 100, 1, 1, device=device)
+        # D step
+        optD.zero_grad(set_to_none=True)
+        fake = G(z)
+        d_loss = bce(D(real), ones) + bce(D(fake.detach()), zeros)
+        d_loss.backward()
+        optD.step()
+        # G step
+        optG.zero_grad(set_to_none=True)
+        g_loss = bce(D(fake), ones)
+        g_loss.backward()
+        optG.step()
+        return d_loss, g_loss
+
+    for _ in range(args.warmup):
+        step()
+    if world > 1:
+        dist.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    if use_cuda:
+        torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    el = time.perf_counter() - t0
+    if world > 1:
+        t = torch.tensor([el], device=device if use_cuda else "cpu",
+                         dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        el = t.item()
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "images/sec (whole job) DCGAN G+D SyncBN",
+            "value": round(bs * world * args.steps / el, 2),
+            "unit": "images/sec",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(1000 * el / args.steps, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {"model": "dcgan64", "per_gpu_batch": bs,
+                       "parallelism": f"dp{world}"},
+        }))
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,131 @@
+#!/usr/bin/env python3
+"""RetinaNet-R50-FPN SyncBN, 2 img/GPU at 800x1333 (BASELINE.json config 5:
+the detection small-batch regime SyncBN exists for).
+
+    python benchmarks/bench_retinanet.py [--steps 10 --warmup 3]
+    torchrun --nproc-per-node 8 benchmarks/bench_retinanet.py ...
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+import torch.distributed as dist
+
+import msbn
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--steps", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--batch-size", type=int, default=2)
+    p.add_argument("--height", type=int, default=800)
+    p.add_argument("--width", type=int, default=1333)
+    p.add_argument("--dtype", type=str, default="bf16",
+                   choices=["bf16", "fp32"])
+    p.add_argument("--local_rank", "--local-rank", type=int,
+                   default=int(os.environ.get("LOCAL_RANK", 0)),
+                   dest="local_rank")
+    args = p.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    use_cuda = torch.cuda.is_available()
+    device = torch.device(f"cuda:{args.local_rank}" if use_cuda else "cpu")
+    if use_cuda:
+        torch.cuda.set_device(device)
+    if world > 1:
+        dist.init_process_group("nccl" if use_cuda else "gloo",
+                                init_method="env://", world_size=world,
+                                rank=rank)
+
+    torch.manual_seed(11)
+    model = msbn.convert_sync_batchnorm(msbn.models.retinanet()).to(device)
+    dtype = torch.bfloat16 if (args.dtype == "bf16" and use_cuda) else \
+        torch.float32
+    if dtype == torch.bfloat16:
+        from bench import cast_bf16_keep_bn_fp32
+
+        model = cast_bf16_keep_bn_fp32(model)
+    if use_cuda:
+        model = model.to(memory_format=torch.channels_last)
+    if world > 1:
+        model = msbn.parallel.DistributedDataParallel(
+            model, device_ids=[args.local_rank] if use_cuda else None,
+            gradient_as_bucket_view=True)
+    opt = torch.optim.SGD(model.parameters(), lr=0.01, momentum=0.9)
+
+    # pad to multiples of 32 for the FPN strides (real detectors pad too)
+    H = (args.height + 31) // 32 * 32
+    W = (args.width + 31) // 32 * 32
+    x = torch.randn(args.batch_size, 3, H, W, device=device, dtype=dtype)
+    if use_cuda:
+        x = x.to(memory_format=torch.channels_last)
+
+    inner = model.module if hasattr(model, "module") else model
+
+    def step():
+        opt.zero_grad(set_to_none=True)
+        if world > 1:
+            # run through the DDP wrapper for bucket hooks
+            cls_outs, box_outs = model(x)
+            loss = x.new_zeros((), dtype=torch.float32)
+            for c, b in zip(cls_outs, box_outs):
+                loss = loss + torch.sigmoid(c.float()).pow(2).mean() \
+                    + b.float().pow(2).mean()
+        else:
+            loss = inner.training_loss(x)
+        loss.backward()
+        opt.step()
+        return loss
+
+    for _ in range(args.warmup):
+        step()
+    if world > 1:
+        dist.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    if use_cuda:
+        torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    el = time.perf_counter() - t0
+    if world > 1:
+        t = torch.tensor([el], device=device if use_cuda else "cpu",
+                         dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        el = t.item()
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "images/sec (whole job) RetinaNet-R50-FPN SyncBN",
+            "value": round(args.batch_size * world * args.steps / el, 2),
+            "unit": "images/sec",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(1000 * el / args.steps, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "dtype": args.dtype,
+            "data": "synthetic",
+            "config": {"model": "retinanet_r50_fpn",
+                       "image": f"3x{H}x{W}",
+                       "per_gpu_batch": args.batch_size,
+                       "parallelism": f"dp{world}"},
+        }))
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -49,20 +49,35 @@ class SyncBatchNormActFunction(torch.autograd.Function):
             input, eps, momentum, running_mean, running_var,
             process_group, world_size,
         )
-        ctx.save_for_backward(input, residual, weight, bias, mean, invstd,
-                              count_sum)
+        # per-channel [scale | shift] computed ONCE and reused by the two
+        # mask-recomputing backward kernels (saves ~100 tiny launches/step
+        # on ResNet-50)
+        coefs = ops.bn_make_coefs(mean, invstd, weight, bias)
+        if coefs is None:
+            ctx.save_for_backward(input, residual, weight, bias, mean, invstd,
+                                  count_sum)
+            ctx.has_coefs = False
+        else:
+            ctx.save_for_backward(input, residual, weight, bias, mean, invstd,
+                                  count_sum, coefs)
+            ctx.has_coefs = True
         ctx.relu = relu
         ctx.process_group = process_group
         ctx.world_size = world_size
         return ops.batch_norm_elemt_act(
-            input, residual, weight, bias, mean, invstd, relu
+            input, residual, weight, bias, mean, invstd, relu, coefs
         )
 
     @staticmethod
     def backward(ctx, grad_output: torch.Tensor):
-        input, residual, weight, bias, mean, invstd, count_sum = (
-            ctx.saved_tensors
-        )
+        if ctx.has_coefs:
+            (input, residual, weight, bias, mean, invstd, count_sum,
+             coefs) = ctx.saved_tensors
+        else:
+            input, residual, weight, bias, mean, invstd, count_sum = (
+                ctx.saved_tensors
+            )
+            coefs = None
         grad_output = _match_layout(grad_output, input)
         relu = ctx.relu
         process_group = ctx.process_group
@@ -76,7 +91,7 @@ class SyncBatchNormActFunction(torch.autograd.Function):
         sum_dy, sum_dy_xmu, grad_weight, grad_bias = (
             ops.batch_norm_backward_reduce_act(
                 grad_output, input, residual, mean, invstd, weight, bias,
-                relu, need_input_g, need_weight_g, need_bias_g,
+                relu, need_input_g, need_weight_g, need_bias_g, coefs,
             )
         )
         grad_input = grad_res = None
@@ -87,7 +102,7 @@ class SyncBatchNormActFunction(torch.autograd.Function):
                 sum_dy, sum_dy_xmu = combined[:C], combined[C:]
             grad_input, grad_res = ops.batch_norm_backward_elemt_act(
                 grad_output, input, residual, mean, invstd, weight, bias,
-                sum_dy, sum_dy_xmu, count_sum, relu, need_res_g,
+                sum_dy, sum_dy_xmu, count_sum, relu, need_res_g, coefs,
             )
         return (
             grad_input if need_input_g else None,

@@ -166,11 +166,18 @@ def batch_norm_backward_elemt(
     )
 
 
+def bn_make_coefs(mean, invstd, weight, bias):
+    """Packed [scale | shift] fp32 per-channel affine (GPU); None on CPU."""
+    if mean.is_cuda:
+        return _require_hip().bn_make_coefs(mean, invstd, weight, bias)
+    return None
+
+
 def batch_norm_elemt_act(input, residual, weight, bias, mean, invstd,
-                         relu: bool):
+                         relu: bool, coefs=None):
     if input.is_cuda:
         return _require_hip().batch_norm_elemt_act(
-            input, residual, weight, bias, mean, invstd, relu
+            input, residual, weight, bias, mean, invstd, relu, coefs
         )
     return _ref.batch_norm_elemt_act(
         input, residual, weight, bias, mean, invstd, relu
@@ -179,11 +186,11 @@ def batch_norm_elemt_act(input, residual, weight, bias, mean, invstd,
 
 def batch_norm_backward_reduce_act(grad_out, input, residual, mean, invstd,
                                    weight, bias, relu_mask, input_g, weight_g,
-                                   bias_g):
+                                   bias_g, coefs=None):
     if input.is_cuda:
         return _require_hip().batch_norm_backward_reduce_act(
             grad_out, input, residual, mean, invstd, weight, bias, relu_mask,
-            input_g, weight_g, bias_g,
+            input_g, weight_g, bias_g, coefs,
         )
     return _ref.batch_norm_backward_reduce_act(
         grad_out, input, residual, mean, invstd, weight, bias, relu_mask,
@@ -193,11 +200,11 @@ def batch_norm_backward_reduce_act(grad_out, input, residual, mean, invstd,
 
 def batch_norm_backward_elemt_act(grad_out, input, residual, mean, invstd,
                                   weight, bias, sum_dy, sum_dy_xmu, count,
-                                  relu_mask, want_res_grad):
+                                  relu_mask, want_res_grad, coefs=None):
     if input.is_cuda:
         dx, dres = _require_hip().batch_norm_backward_elemt_act(
             grad_out, input, residual, mean, invstd, weight, bias, sum_dy,
-            sum_dy_xmu, count, relu_mask, want_res_grad,
+            sum_dy_xmu, count, relu_mask, want_res_grad, coefs,
         )
         return dx, (dres if want_res_grad else None)
     return _ref.batch_norm_backward_elemt_act(
@@ -209,6 +216,7 @@ def batch_norm_backward_elemt_act(grad_out, input, residual, mean, invstd,
 __all__ = [
     "hip_available",
     "batch_norm_elemt_act",
+    "bn_make_coefs",
     "batch_norm_backward_reduce_act",
     "batch_norm_backward_elemt_act",
     "batch_norm_stats",

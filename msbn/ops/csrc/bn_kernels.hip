@@ -1113,17 +1113,26 @@ at::Tensor make_fwd_coefs(const at::Tensor& mean, const at::Tensor& invstd,
   }()
 }  // namespace
 
+at::Tensor bn_make_coefs(const at::Tensor& mean, const at::Tensor& invstd,
+                         const c10::optional<at::Tensor>& weight,
+                         const c10::optional<at::Tensor>& bias) {
+  return make_fwd_coefs(mean, invstd, weight, bias, mean.numel());
+}
+
 at::Tensor batch_norm_elemt_act(const at::Tensor& input,
                                 const c10::optional<at::Tensor>& residual,
                                 const c10::optional<at::Tensor>& weight,
                                 const c10::optional<at::Tensor>& bias,
                                 const at::Tensor& mean,
-                                const at::Tensor& invstd, bool relu) {
+                                const at::Tensor& invstd, bool relu,
+                                const c10::optional<at::Tensor>& coefs_in) {
   const Layout L = get_layout(input);
   auto out = at::empty_like(input);
   if (input.numel() == 0) return out;
   auto stream = cur_stream();
-  auto coefs = make_fwd_coefs(mean, invstd, weight, bias, L.C);
+  auto coefs = coefs_in.has_value()
+                   ? *coefs_in
+                   : make_fwd_coefs(mean, invstd, weight, bias, L.C);
   float* scale = coefs.data_ptr<float>();
   float* shift = scale + L.C;
   const bool has_res = residual.has_value();
@@ -1172,7 +1181,7 @@ at::Tensor batch_norm_elemt(const at::Tensor& input,
                             double eps) {
   (void)eps;  // invstd already folds eps
   return batch_norm_elemt_act(input, c10::nullopt, weight, bias, mean, invstd,
-                              false);
+                              false, c10::nullopt);
 }
 
 std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor>
@@ -1181,7 +1190,7 @@ batch_norm_backward_reduce_act(
     const c10::optional<at::Tensor>& residual, const at::Tensor& mean,
     const at::Tensor& invstd, const c10::optional<at::Tensor>& weight,
     const c10::optional<at::Tensor>& bias, bool relu_mask, bool input_g,
-    bool weight_g, bool bias_g) {
+    bool weight_g, bool bias_g, const c10::optional<at::Tensor>& coefs_in) {
   const Layout L = get_layout(input);
   TORCH_CHECK(grad_out.sizes() == input.sizes() &&
                   grad_out.strides() == input.strides(),
@@ -1207,7 +1216,9 @@ batch_norm_backward_reduce_act(
   float* shift = nullptr;
   at::Tensor coefs;
   if (relu_mask) {
-    coefs = make_fwd_coefs(mean, invstd, weight, bias, L.C);
+    coefs = coefs_in.has_value()
+                ? *coefs_in
+                : make_fwd_coefs(mean, invstd, weight, bias, L.C);
     scale = coefs.data_ptr<float>();
     shift = scale + L.C;
   }
@@ -1291,7 +1302,8 @@ batch_norm_backward_reduce(const at::Tensor& grad_out, const at::Tensor& input,
                            bool weight_g, bool bias_g) {
   return batch_norm_backward_reduce_act(grad_out, input, c10::nullopt, mean,
                                         invstd, weight, c10::nullopt, false,
-                                        input_g, weight_g, bias_g);
+                                        input_g, weight_g, bias_g,
+                                        c10::nullopt);
 }
 
 std::tuple<at::Tensor, at::Tensor> batch_norm_backward_elemt_act(
@@ -1300,7 +1312,7 @@ std::tuple<at::Tensor, at::Tensor> batch_norm_backward_elemt_act(
     const at::Tensor& invstd, const c10::optional<at::Tensor>& weight,
     const c10::optional<at::Tensor>& bias, const at::Tensor& sum_dy,
     const at::Tensor& sum_dy_xmu, const at::Tensor& count, bool relu_mask,
-    bool want_res_grad) {
+    bool want_res_grad, const c10::optional<at::Tensor>& coefs_in) {
   const Layout L = get_layout(input);
   auto dx = at::empty_like(grad_out);
   at::Tensor dres;
@@ -1338,7 +1350,9 @@ std::tuple<at::Tensor, at::Tensor> batch_norm_backward_elemt_act(
   float* shift = nullptr;
   at::Tensor fcoefs;
   if (relu_mask) {
-    fcoefs = make_fwd_coefs(mean, invstd, weight, bias, L.C);
+    fcoefs = coefs_in.has_value()
+                 ? *coefs_in
+                 : make_fwd_coefs(mean, invstd, weight, bias, L.C);
     scale = fcoefs.data_ptr<float>();
     shift = scale + L.C;
   }
@@ -1390,7 +1404,7 @@ at::Tensor batch_norm_backward_elemt(
     const at::Tensor& count) {
   return std::get<0>(batch_norm_backward_elemt_act(
       grad_out, input, c10::nullopt, mean, invstd, weight, c10::nullopt,
-      sum_dy, sum_dy_xmu, count, false, false));
+      sum_dy, sum_dy_xmu, count, false, false, c10::nullopt));
 }
 
 }  // namespace msbn

@@ -60,12 +60,19 @@ at::Tensor batch_norm_backward_elemt(
 // ReLU gate in-kernel from (x, scale, shift[, residual]) — no mask tensor and
 // no separate clamp/add/threshold_backward kernels or their full-tensor
 // round trips.
+// coefs = packed [scale(C) | shift(C)] fp32 (bn_make_coefs); pass it to the
+// act ops to avoid recomputing the per-channel affine in each of them.
+at::Tensor bn_make_coefs(const at::Tensor& mean, const at::Tensor& invstd,
+                         const c10::optional<at::Tensor>& weight,
+                         const c10::optional<at::Tensor>& bias);
+
 at::Tensor batch_norm_elemt_act(const at::Tensor& input,
                                 const c10::optional<at::Tensor>& residual,
                                 const c10::optional<at::Tensor>& weight,
                                 const c10::optional<at::Tensor>& bias,
                                 const at::Tensor& mean,
-                                const at::Tensor& invstd, bool relu);
+                                const at::Tensor& invstd, bool relu,
+                                const c10::optional<at::Tensor>& coefs_in);
 
 std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor>
 batch_norm_backward_reduce_act(
@@ -73,7 +80,7 @@ batch_norm_backward_reduce_act(
     const c10::optional<at::Tensor>& residual, const at::Tensor& mean,
     const at::Tensor& invstd, const c10::optional<at::Tensor>& weight,
     const c10::optional<at::Tensor>& bias, bool relu_mask, bool input_g,
-    bool weight_g, bool bias_g);
+    bool weight_g, bool bias_g, const c10::optional<at::Tensor>& coefs_in);
 
 // returns (grad_input, grad_residual-or-undefined)
 std::tuple<at::Tensor, at::Tensor> batch_norm_backward_elemt_act(
@@ -82,6 +89,6 @@ std::tuple<at::Tensor, at::Tensor> batch_norm_backward_elemt_act(
     const at::Tensor& invstd, const c10::optional<at::Tensor>& weight,
     const c10::optional<at::Tensor>& bias, const at::Tensor& sum_dy,
     const at::Tensor& sum_dy_xmu, const at::Tensor& count, bool relu_mask,
-    bool want_res_grad);
+    bool want_res_grad, const c10::optional<at::Tensor>& coefs_in);
 
 }  // namespace msbn

@@ -33,20 +33,24 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("grad_out"), py::arg("input"), py::arg("mean"),
         py::arg("invstd"), py::arg("weight"), py::arg("sum_dy"),
         py::arg("sum_dy_xmu"), py::arg("count"));
+  m.def("bn_make_coefs", &msbn::bn_make_coefs, py::arg("mean"),
+        py::arg("invstd"), py::arg("weight"), py::arg("bias"));
   m.def("batch_norm_elemt_act", &msbn::batch_norm_elemt_act, py::arg("input"),
         py::arg("residual"), py::arg("weight"), py::arg("bias"),
-        py::arg("mean"), py::arg("invstd"), py::arg("relu"));
+        py::arg("mean"), py::arg("invstd"), py::arg("relu"),
+        py::arg("coefs") = py::none());
   m.def("batch_norm_backward_reduce_act",
         &msbn::batch_norm_backward_reduce_act, py::arg("grad_out"),
         py::arg("input"), py::arg("residual"), py::arg("mean"),
         py::arg("invstd"), py::arg("weight"), py::arg("bias"),
         py::arg("relu_mask"), py::arg("input_g"), py::arg("weight_g"),
-        py::arg("bias_g"));
+        py::arg("bias_g"), py::arg("coefs") = py::none());
   m.def("batch_norm_backward_elemt_act", &msbn::batch_norm_backward_elemt_act,
         py::arg("grad_out"), py::arg("input"), py::arg("residual"),
         py::arg("mean"), py::arg("invstd"), py::arg("weight"), py::arg("bias"),
         py::arg("sum_dy"), py::arg("sum_dy_xmu"), py::arg("count"),
-        py::arg("relu_mask"), py::arg("want_res_grad"));
+        py::arg("relu_mask"), py::arg("want_res_grad"),
+        py::arg("coefs") = py::none());
 
   // ---- DDP machinery ----
   m.def("compute_bucket_assignment_by_size",
