@@ -264,9 +264,12 @@ class DistributedDataParallel(Module):
 
     @contextmanager
     def join(self, divide_by_initial_world_size: bool = True, enable: bool = True):
-        """Minimal uneven-input helper: msbn SyncBN tolerates empty inputs and
-        the reducer requires symmetric iteration counts; feed zero-size batches
-        on exhausted ranks (see tests/test_ddp_cpu.py::test_uneven_inputs)."""
+        """Uneven-input support lives in msbn.parallel.run_with_join: exhausted
+        ranks step on empty batches (SyncBN masks zero-count stats in-kernel;
+        the reducer all-reduces zero grads), keeping every collective matched.
+        This context is a compatibility shim for code structured around stock
+        DDP.join(); inside it the caller must keep iteration counts symmetric
+        or drive the loop with run_with_join."""
         yield
 
     def _get_ddp_logging_data(self):

@@ -255,6 +255,45 @@ def test_checkpoint_roundtrip(tmp_path):
 
 
 # --------------------------------------------------------------------------
+def _join_uneven_body(rank):
+    """run_with_join: rank 0 has 4 batches, rank 1 has 2; both must finish
+    with identical (averaged) parameters and no hang."""
+    import msbn
+    from msbn.parallel import run_with_join
+
+    torch.manual_seed(3)
+    net = msbn.convert_sync_batchnorm(msbn.models.SimpleCNN(width=8))
+    net = msbn.parallel.DistributedDataParallel(net)
+    opt = torch.optim.SGD(net.parameters(), lr=0.05)
+
+    n_batches = 4 if rank == 0 else 2
+    data = [torch.randn(2, 3, 8, 8, generator=torch.Generator().manual_seed(
+        rank * 100 + i)) for i in range(n_batches)]
+
+    def step(x):
+        opt.zero_grad(set_to_none=True)
+        out = net(x)
+        loss = out.float().pow(2).sum() * (1.0 / max(1, x.shape[0]))
+        loss.backward()
+        opt.step()
+
+    real = run_with_join(
+        net, data, step_fn=step,
+        make_empty_batch=lambda: torch.randn(0, 3, 8, 8),
+    )
+    assert real == n_batches
+    # parameters identical across ranks afterwards
+    flat = torch.cat([p.detach().flatten() for p in net.module.parameters()])
+    flat0 = flat.clone()
+    dist.broadcast(flat0, src=0)
+    assert torch.allclose(flat, flat0, atol=1e-6)
+
+
+def test_join_uneven_inputs(tmp_path):
+    _spawn("_join_uneven_body", tmp_path)
+
+
+# --------------------------------------------------------------------------
 def _rebuild_buckets_body(rank):
     import msbn
 
