@@ -57,7 +57,7 @@ def main():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=30)
     p.add_argument("--warmup", type=int, default=10)
-    p.add_argument("--batch-size", type=int, default=256,
+    p.add_argument("--batch-size", type=int, default=512,
                    help="per-GPU batch (weak scaling)")
     p.add_argument("--model", type=str, default="resnet50")
     # channels_last default: fastest measured path on gfx950 (MIOpen igemm

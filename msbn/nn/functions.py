@@ -66,12 +66,18 @@ def compute_sync_stats(
     running_var: Optional[torch.Tensor],
     process_group,
     world_size: int,
+    weight: Optional[torch.Tensor] = None,
+    bias: Optional[torch.Tensor] = None,
+    want_coefs: bool = False,
 ):
     """Local packed moments -> all_gather over RCCL/xGMI (S5) -> counts-weighted
-    combine with in-kernel zero-count masking + running-stats update.
-    Returns (mean, invstd, count_sum[1])."""
+    combine with in-kernel zero-count masking + running-stats update.  The
+    combine/finalize kernel also emits the per-channel [scale|shift] coefs
+    when requested (one launch for the whole per-layer channel math).
+    Returns (mean, invstd, count_sum[1], coefs-or-None)."""
     C = int(input.shape[1])
     local_count = input.numel() // C if C > 0 else 0
+    on_gpu = input.is_cuda
 
     if world_size > 1:
         packed = torch.empty(2 * C + 1, dtype=torch.float32, device=input.device)
@@ -83,36 +89,59 @@ def compute_sync_stats(
         packed_all = torch.empty(
             (world_size, 2 * C + 1), dtype=torch.float32, device=input.device
         )
+        from msbn.utils.logging import comm_log
+
         if _is_nccl_like(process_group):
             dist.all_gather_into_tensor(packed_all, packed, group=process_group)
         else:
             chunks = list(packed_all.unbind(0))
             dist.all_gather(chunks, packed, group=process_group)
             packed_all = torch.stack(chunks, dim=0)
+        comm_log.record("all_gather", packed.numel() * 4, f"syncbn C={C}")
 
+        if on_gpu and want_coefs:
+            from msbn.ops import _require_hip
+
+            mean, invstd, count_sum, coefs = (
+                _require_hip().batch_norm_gather_stats_packed_coefs(
+                    packed_all, running_mean, running_var, momentum, eps,
+                    weight, bias, True,
+                )
+            )
+            return mean, invstd, count_sum, coefs
         mean, invstd, count_sum = ops.batch_norm_gather_stats_packed(
             input, packed_all, running_mean, running_var, momentum, eps
         )
-    else:
-        mean, invstd = ops.batch_norm_stats(input, eps)
-        count_sum = torch.full(
-            (1,), float(local_count), dtype=torch.float32, device=input.device
+        return mean, invstd, count_sum, None
+
+    if on_gpu and local_count > 0:
+        from msbn.ops import _require_hip
+
+        mean, invstd, count_sum, coefs = _require_hip().batch_norm_stats_local(
+            input, eps, running_mean, running_var, momentum, weight, bias,
+            want_coefs,
         )
-        if running_mean is not None and local_count > 0:
-            with torch.no_grad():
-                var = invstd.to(torch.float32).pow(-2) - eps
-                unbiased = (
-                    var * (local_count / (local_count - 1.0))
-                    if local_count > 1
-                    else var
-                )
-                running_mean.mul_(1 - momentum).add_(
-                    mean.to(running_mean.dtype), alpha=momentum
-                )
-                running_var.mul_(1 - momentum).add_(
-                    unbiased.to(running_var.dtype), alpha=momentum
-                )
-    return mean, invstd, count_sum
+        return mean, invstd, count_sum, (coefs if want_coefs else None)
+
+    mean, invstd = ops.batch_norm_stats(input, eps)
+    count_sum = torch.full(
+        (1,), float(local_count), dtype=torch.float32, device=input.device
+    )
+    if running_mean is not None and local_count > 0:
+        with torch.no_grad():
+            var = invstd.to(torch.float32).pow(-2) - eps
+            unbiased = (
+                var * (local_count / (local_count - 1.0))
+                if local_count > 1
+                else var
+            )
+            running_mean.mul_(1 - momentum).add_(
+                mean.to(running_mean.dtype), alpha=momentum
+            )
+            running_var.mul_(1 - momentum).add_(
+                unbiased.to(running_var.dtype), alpha=momentum
+            )
+    return mean, invstd, count_sum, None
 
 
 class SyncBatchNormFunction(torch.autograd.Function):
@@ -136,9 +165,9 @@ class SyncBatchNormFunction(torch.autograd.Function):
         C = int(input.shape[1])
         local_count = input.numel() // C if C > 0 else 0
 
-        mean, invstd, count_sum = compute_sync_stats(
+        mean, invstd, count_sum, coefs = compute_sync_stats(
             input, eps, momentum, running_mean, running_var,
-            process_group, world_size,
+            process_group, world_size, weight, bias, want_coefs=True,
         )
 
         ctx.save_for_backward(input, weight, mean, invstd, count_sum)
@@ -147,7 +176,9 @@ class SyncBatchNormFunction(torch.autograd.Function):
 
         if local_count == 0:
             return torch.empty_like(input)
-        return ops.batch_norm_elemt(input, weight, bias, mean, invstd, eps)
+        return ops.batch_norm_elemt_act(
+            input, None, weight, bias, mean, invstd, False, coefs
+        )
 
     @staticmethod
     def backward(ctx, grad_output: torch.Tensor):
@@ -172,6 +203,9 @@ class SyncBatchNormFunction(torch.autograd.Function):
                     dist.all_reduce(
                         combined, dist.ReduceOp.SUM, group=process_group
                     )
+                    from msbn.utils.logging import comm_log
+                    comm_log.record("all_reduce", combined.numel() * 4,
+                                    f"syncbn bwd C={C}")
                     sum_dy, sum_dy_xmu = combined[:C], combined[C:]
                 grad_input = ops.batch_norm_backward_elemt(
                     grad_output, input, mean, invstd, weight,

@@ -45,14 +45,13 @@ class SyncBatchNormActFunction(torch.autograd.Function):
         if bias is not None:
             bias = bias.contiguous()
 
-        mean, invstd, count_sum = compute_sync_stats(
+        # coefs ([scale|shift]) are emitted by the finalize/gather kernel in
+        # the same launch and reused by forward elemt AND the two
+        # mask-recomputing backward kernels.
+        mean, invstd, count_sum, coefs = compute_sync_stats(
             input, eps, momentum, running_mean, running_var,
-            process_group, world_size,
+            process_group, world_size, weight, bias, want_coefs=True,
         )
-        # per-channel [scale | shift] computed ONCE and reused by the two
-        # mask-recomputing backward kernels (saves ~100 tiny launches/step
-        # on ResNet-50)
-        coefs = ops.bn_make_coefs(mean, invstd, weight, bias)
         if coefs is None:
             ctx.save_for_backward(input, residual, weight, bias, mean, invstd,
                                   count_sum)
@@ -64,6 +63,8 @@ class SyncBatchNormActFunction(torch.autograd.Function):
         ctx.relu = relu
         ctx.process_group = process_group
         ctx.world_size = world_size
+        if input.numel() == 0:
+            return torch.empty_like(input)
         return ops.batch_norm_elemt_act(
             input, residual, weight, bias, mean, invstd, relu, coefs
         )
@@ -88,6 +89,21 @@ class SyncBatchNormActFunction(torch.autograd.Function):
         need_bias_g = bias is not None and ctx.needs_input_grad[3]
 
         C = int(input.shape[1])
+        if input.numel() == 0:
+            # empty-input rank (join): contribute zeros, keep peers unblocked
+            if world_size > 1 and (need_input_g or need_res_g):
+                combined = torch.zeros(2 * C, dtype=torch.float32,
+                                       device=grad_output.device)
+                dist.all_reduce(combined, dist.ReduceOp.SUM,
+                                group=process_group)
+            zg = torch.empty_like(grad_output)
+            return (
+                zg if need_input_g else None,
+                torch.empty_like(grad_output) if need_res_g else None,
+                torch.zeros_like(weight) if need_weight_g else None,
+                torch.zeros_like(bias) if need_bias_g else None,
+                None, None, None, None, None, None, None,
+            )
         sum_dy, sum_dy_xmu, grad_weight, grad_bias = (
             ops.batch_norm_backward_reduce_act(
                 grad_output, input, residual, mean, invstd, weight, bias,
@@ -99,6 +115,9 @@ class SyncBatchNormActFunction(torch.autograd.Function):
             if world_size > 1:
                 combined = torch.cat([sum_dy, sum_dy_xmu])
                 dist.all_reduce(combined, dist.ReduceOp.SUM, group=process_group)
+                from msbn.utils.logging import comm_log
+                comm_log.record("all_reduce", combined.numel() * 4,
+                                f"syncbn bwd C={C}")
                 sum_dy, sum_dy_xmu = combined[:C], combined[C:]
             grad_input, grad_res = ops.batch_norm_backward_elemt_act(
                 grad_output, input, residual, mean, invstd, weight, bias,

@@ -175,14 +175,17 @@ __global__ void bn_stats_partial_nhwc(const T* __restrict__ x,
 // lane-strided loads + wave64 shuffle reduce); nchunks may be in the
 // thousands for small-C layers without serializing.
 // =====================================================================
-template <typename RT>
+template <typename RT, typename WT>
 __global__ void bn_stats_finalize(const double* __restrict__ ws, int nchunks,
                                   int64_t C, double count, float eps,
                                   float* __restrict__ mean,
                                   float* __restrict__ invstd,
                                   float* __restrict__ count_out,
                                   RT* __restrict__ rmean, RT* __restrict__ rvar,
-                                  float momentum) {
+                                  float momentum, const WT* __restrict__ w,
+                                  const WT* __restrict__ bws,
+                                  float* __restrict__ scale_out,
+                                  float* __restrict__ shift_out) {
   const int lane = threadIdx.x & (MSBN_WAVE - 1);
   const int wid = threadIdx.x >> 6;
   const int64_t c = (int64_t)blockIdx.x * (blockDim.x >> 6) + wid;
@@ -197,8 +200,9 @@ __global__ void bn_stats_finalize(const double* __restrict__ ws, int nchunks,
   const double m = a / count;
   double var = b / count - m * m;
   var = var > 0.0 ? var : 0.0;
+  const float istd = (float)rsqrt(var + (double)eps);
   mean[c] = (float)m;
-  invstd[c] = (float)rsqrt(var + (double)eps);
+  invstd[c] = istd;
   if (c == 0 && count_out != nullptr) count_out[0] = (float)count;
   if (rmean != nullptr) {
     const double unbiased = count > 1.0 ? var * (count / (count - 1.0)) : var;
@@ -206,19 +210,28 @@ __global__ void bn_stats_finalize(const double* __restrict__ ws, int nchunks,
     rvar[c] =
         from_f<RT>((1.f - momentum) * to_f(rvar[c]) + momentum * (float)unbiased);
   }
+  if (scale_out != nullptr) {
+    const float sc = istd * (w != nullptr ? to_f(w[c]) : 1.f);
+    scale_out[c] = sc;
+    shift_out[c] = -(float)m * sc + (bws != nullptr ? to_f(bws[c]) : 0.f);
+  }
 }
 
 // =====================================================================
 // gather: combine W ranks' packed [mean | invstd | count] rows.
 // Zero-count ranks masked HERE (device-side; no host sync).
 // =====================================================================
-template <typename RT>
+template <typename RT, typename WT>
 __global__ void bn_gather_stats(const float* __restrict__ packed_all, int W,
                                 int64_t C, float eps, float momentum,
                                 float* __restrict__ mean,
                                 float* __restrict__ invstd,
                                 float* __restrict__ count_out,
-                                RT* __restrict__ rmean, RT* __restrict__ rvar) {
+                                RT* __restrict__ rmean, RT* __restrict__ rvar,
+                                const WT* __restrict__ wpar,
+                                const WT* __restrict__ bpar,
+                                float* __restrict__ scale_out,
+                                float* __restrict__ shift_out) {
   const int64_t c = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
   const int64_t row = 2 * C + 1;
@@ -234,25 +247,29 @@ __global__ void bn_gather_stats(const float* __restrict__ packed_all, int W,
       ex2 += cnt * (var + m * m);
     }
   }
-  if (n_tot == 0.0) {
-    mean[c] = 0.f;
-    invstd[c] = 0.f;
-    if (c == 0 && count_out != nullptr) count_out[0] = 0.f;
-    return;
+  float m_out = 0.f, istd_out = 0.f;
+  if (n_tot > 0.0) {
+    const double m_g = m_acc / n_tot;
+    double var_g = ex2 / n_tot - m_g * m_g;
+    var_g = var_g > 0.0 ? var_g : 0.0;
+    m_out = (float)m_g;
+    istd_out = (float)rsqrt(var_g + (double)eps);
+    if (rmean != nullptr) {
+      const double unbiased =
+          n_tot > 1.0 ? var_g * (n_tot / (n_tot - 1.0)) : var_g;
+      rmean[c] =
+          from_f<RT>((1.f - momentum) * to_f(rmean[c]) + momentum * (float)m_g);
+      rvar[c] = from_f<RT>((1.f - momentum) * to_f(rvar[c]) +
+                           momentum * (float)unbiased);
+    }
   }
-  const double m_g = m_acc / n_tot;
-  double var_g = ex2 / n_tot - m_g * m_g;
-  var_g = var_g > 0.0 ? var_g : 0.0;
-  mean[c] = (float)m_g;
-  invstd[c] = (float)rsqrt(var_g + (double)eps);
+  mean[c] = m_out;
+  invstd[c] = istd_out;
   if (c == 0 && count_out != nullptr) count_out[0] = (float)n_tot;
-  if (rmean != nullptr) {
-    const double unbiased =
-        n_tot > 1.0 ? var_g * (n_tot / (n_tot - 1.0)) : var_g;
-    rmean[c] =
-        from_f<RT>((1.f - momentum) * to_f(rmean[c]) + momentum * (float)m_g);
-    rvar[c] = from_f<RT>((1.f - momentum) * to_f(rvar[c]) +
-                         momentum * (float)unbiased);
+  if (scale_out != nullptr) {
+    const float sc = istd_out * (wpar != nullptr ? to_f(wpar[c]) : 1.f);
+    scale_out[c] = sc;
+    shift_out[c] = -m_out * sc + (bpar != nullptr ? to_f(bpar[c]) : 0.f);
   }
 }
 
@@ -923,13 +940,39 @@ hipStream_t cur_stream() {
 
 // Launch the two-stage stats reduction; writes mean/invstd/count into the
 // given fp32 pointers (which may alias a packed buffer).
+#define MSBN_DISPATCH_WTYPE(TYPE, NAME, ...)                                \
+  [&] {                                                                     \
+    switch (TYPE) {                                                         \
+      case at::kFloat: {                                                    \
+        using wt_t = float;                                                 \
+        return __VA_ARGS__();                                               \
+      }                                                                     \
+      case at::kBFloat16: {                                                 \
+        using wt_t = __hip_bfloat16;                                        \
+        return __VA_ARGS__();                                               \
+      }                                                                     \
+      case at::kHalf: {                                                     \
+        using wt_t = __half;                                                \
+        return __VA_ARGS__();                                               \
+      }                                                                     \
+      default:                                                              \
+        TORCH_CHECK(false, NAME, ": unsupported weight dtype ", TYPE);      \
+    }                                                                       \
+  }()
+
 void stats_into(const at::Tensor& input, double eps, float* mean_p,
                 float* invstd_p, float* count_p, const at::Tensor* rmean,
-                const at::Tensor* rvar, double momentum) {
+                const at::Tensor* rvar, double momentum,
+                const at::Tensor* wpar = nullptr,
+                const at::Tensor* bpar = nullptr, float* scale_p = nullptr,
+                float* shift_p = nullptr) {
   const Layout L = get_layout(input);
   const int64_t count = L.rows;
   auto stream = cur_stream();
   const auto rstat_type = rmean ? rmean->scalar_type() : at::kFloat;
+  const auto w_type = wpar ? wpar->scalar_type()
+                     : bpar ? bpar->scalar_type()
+                            : at::kFloat;
 
   MSBN_DISPATCH_FLOAT_TYPES(input.scalar_type(), "batch_norm_stats", [&] {
     const native_t* x =
@@ -974,14 +1017,21 @@ void stats_into(const at::Tensor& input, double eps, float* mean_p,
     }
     const int fgrid = (int)cdiv(L.C, kFinalizeWavesPerBlock);
     MSBN_DISPATCH_RSTAT(rstat_type, "batch_norm_stats", [&] {
-      rstat_t* rm = rmean ? reinterpret_cast<rstat_t*>(rmean->data_ptr())
-                          : nullptr;
-      rstat_t* rv = rvar ? reinterpret_cast<rstat_t*>(rvar->data_ptr())
-                         : nullptr;
-      hipLaunchKernelGGL((bn_stats_finalize<rstat_t>), dim3(fgrid),
-                         dim3(MSBN_BLOCK), 0, stream, ws.data_ptr<double>(),
-                         nchunks, L.C, (double)count, (float)eps, mean_p,
-                         invstd_p, count_p, rm, rv, (float)momentum);
+      MSBN_DISPATCH_WTYPE(w_type, "batch_norm_stats", [&] {
+        rstat_t* rm = rmean ? reinterpret_cast<rstat_t*>(rmean->data_ptr())
+                            : nullptr;
+        rstat_t* rv = rvar ? reinterpret_cast<rstat_t*>(rvar->data_ptr())
+                           : nullptr;
+        const wt_t* wp =
+            wpar ? reinterpret_cast<const wt_t*>(wpar->data_ptr()) : nullptr;
+        const wt_t* bp =
+            bpar ? reinterpret_cast<const wt_t*>(bpar->data_ptr()) : nullptr;
+        hipLaunchKernelGGL((bn_stats_finalize<rstat_t, wt_t>), dim3(fgrid),
+                           dim3(MSBN_BLOCK), 0, stream, ws.data_ptr<double>(),
+                           nchunks, L.C, (double)count, (float)eps, mean_p,
+                           invstd_p, count_p, rm, rv, (float)momentum, wp, bp,
+                           scale_p, shift_p);
+      });
     });
   });
 }
@@ -1022,9 +1072,12 @@ void batch_norm_stats_packed(const at::Tensor& input, double eps,
   stats_into(input, eps, p, p + C, p + 2 * C, nullptr, nullptr, 0.0);
 }
 
-std::tuple<at::Tensor, at::Tensor, at::Tensor> batch_norm_gather_stats_packed(
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor>
+batch_norm_gather_stats_packed_coefs(
     const at::Tensor& packed_all, const c10::optional<at::Tensor>& running_mean,
-    const c10::optional<at::Tensor>& running_var, double momentum, double eps) {
+    const c10::optional<at::Tensor>& running_var, double momentum, double eps,
+    const c10::optional<at::Tensor>& weight,
+    const c10::optional<at::Tensor>& bias, bool want_coefs) {
   TORCH_CHECK(packed_all.dim() == 2 && packed_all.is_contiguous() &&
                   packed_all.scalar_type() == at::kFloat,
               "packed_all must be contiguous fp32 [W, 2C+1]");
@@ -1034,25 +1087,90 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> batch_norm_gather_stats_packed(
   auto mean = at::empty({C}, opts);
   auto invstd = at::empty({C}, opts);
   auto count_sum = at::empty({1}, opts);
+  at::Tensor coefs;
+  float* scale_p = nullptr;
+  float* shift_p = nullptr;
+  if (want_coefs) {
+    coefs = at::empty({2 * C}, opts);
+    scale_p = coefs.data_ptr<float>();
+    shift_p = scale_p + C;
+  }
   auto stream = cur_stream();
   const auto rstat_type =
       running_mean.has_value() ? running_mean->scalar_type() : at::kFloat;
+  const auto w_type = weight.has_value() ? weight->scalar_type()
+                      : bias.has_value() ? bias->scalar_type()
+                                         : at::kFloat;
   const int fgrid = (int)cdiv(C, MSBN_BLOCK);
   MSBN_DISPATCH_RSTAT(rstat_type, "gather_stats", [&] {
-    rstat_t* rm = running_mean.has_value()
-                      ? reinterpret_cast<rstat_t*>(running_mean->data_ptr())
-                      : nullptr;
-    rstat_t* rv = running_var.has_value()
-                      ? reinterpret_cast<rstat_t*>(running_var->data_ptr())
-                      : nullptr;
-    hipLaunchKernelGGL((bn_gather_stats<rstat_t>), dim3(fgrid),
-                       dim3(MSBN_BLOCK), 0, stream,
-                       packed_all.data_ptr<float>(), W, C, (float)eps,
-                       (float)momentum, mean.data_ptr<float>(),
-                       invstd.data_ptr<float>(), count_sum.data_ptr<float>(),
-                       rm, rv);
+    MSBN_DISPATCH_WTYPE(w_type, "gather_stats", [&] {
+      rstat_t* rm = running_mean.has_value()
+                        ? reinterpret_cast<rstat_t*>(running_mean->data_ptr())
+                        : nullptr;
+      rstat_t* rv = running_var.has_value()
+                        ? reinterpret_cast<rstat_t*>(running_var->data_ptr())
+                        : nullptr;
+      const wt_t* wp =
+          weight.has_value()
+              ? reinterpret_cast<const wt_t*>(weight->data_ptr())
+              : nullptr;
+      const wt_t* bp = bias.has_value()
+                           ? reinterpret_cast<const wt_t*>(bias->data_ptr())
+                           : nullptr;
+      hipLaunchKernelGGL((bn_gather_stats<rstat_t, wt_t>), dim3(fgrid),
+                         dim3(MSBN_BLOCK), 0, stream,
+                         packed_all.data_ptr<float>(), W, C, (float)eps,
+                         (float)momentum, mean.data_ptr<float>(),
+                         invstd.data_ptr<float>(), count_sum.data_ptr<float>(),
+                         rm, rv, wp, bp, scale_p, shift_p);
+    });
   });
-  return {mean, invstd, count_sum};
+  return {mean, invstd, count_sum, coefs};
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> batch_norm_gather_stats_packed(
+    const at::Tensor& packed_all, const c10::optional<at::Tensor>& running_mean,
+    const c10::optional<at::Tensor>& running_var, double momentum, double eps) {
+  auto r = batch_norm_gather_stats_packed_coefs(
+      packed_all, running_mean, running_var, momentum, eps, c10::nullopt,
+      c10::nullopt, false);
+  return {std::get<0>(r), std::get<1>(r), std::get<2>(r)};
+}
+
+// Fused local (world_size==1) stats: partial+finalize with running-stats
+// update AND [scale|shift] coefs in ONE finalize launch.
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor>
+batch_norm_stats_local(const at::Tensor& input, double eps,
+                       const c10::optional<at::Tensor>& running_mean,
+                       const c10::optional<at::Tensor>& running_var,
+                       double momentum,
+                       const c10::optional<at::Tensor>& weight,
+                       const c10::optional<at::Tensor>& bias,
+                       bool want_coefs) {
+  const int64_t C = input.size(1);
+  auto opts = input.options().dtype(at::kFloat);
+  auto mean = at::empty({C}, opts);
+  auto invstd = at::empty({C}, opts);
+  auto count_sum = at::empty({1}, opts);
+  at::Tensor coefs;
+  float* scale_p = nullptr;
+  float* shift_p = nullptr;
+  if (want_coefs) {
+    coefs = at::empty({2 * C}, opts);
+    scale_p = coefs.data_ptr<float>();
+    shift_p = scale_p + C;
+  }
+  TORCH_CHECK(input.numel() > 0, "batch_norm_stats_local: empty input");
+  const at::Tensor* rm =
+      running_mean.has_value() ? &running_mean.value() : nullptr;
+  const at::Tensor* rv =
+      running_var.has_value() ? &running_var.value() : nullptr;
+  const at::Tensor* wp = weight.has_value() ? &weight.value() : nullptr;
+  const at::Tensor* bp = bias.has_value() ? &bias.value() : nullptr;
+  stats_into(input, eps, mean.data_ptr<float>(), invstd.data_ptr<float>(),
+             count_sum.data_ptr<float>(), rm, rv, momentum, wp, bp, scale_p,
+             shift_p);
+  return {mean, invstd, count_sum, coefs};
 }
 
 std::tuple<at::Tensor, at::Tensor> batch_norm_gather_stats_with_counts(

@@ -31,6 +31,24 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> batch_norm_gather_stats_packed(
     const at::Tensor& packed_all, const c10::optional<at::Tensor>& running_mean,
     const c10::optional<at::Tensor>& running_var, double momentum, double eps);
 
+// As above but ALSO emits the per-channel [scale | shift] coefs (one launch).
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor>
+batch_norm_gather_stats_packed_coefs(
+    const at::Tensor& packed_all, const c10::optional<at::Tensor>& running_mean,
+    const c10::optional<at::Tensor>& running_var, double momentum, double eps,
+    const c10::optional<at::Tensor>& weight,
+    const c10::optional<at::Tensor>& bias, bool want_coefs);
+
+// Fused local (world_size==1) stats: running-stats update + coefs emitted by
+// the finalize kernel -> (mean, invstd, count_sum[1], coefs-or-undefined).
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor>
+batch_norm_stats_local(const at::Tensor& input, double eps,
+                       const c10::optional<at::Tensor>& running_mean,
+                       const c10::optional<at::Tensor>& running_var,
+                       double momentum,
+                       const c10::optional<at::Tensor>& weight,
+                       const c10::optional<at::Tensor>& bias, bool want_coefs);
+
 // y = (x - mean) * invstd * weight + bias (scale/shift precomputed per channel).
 at::Tensor batch_norm_elemt(const at::Tensor& input,
                             const c10::optional<at::Tensor>& weight,

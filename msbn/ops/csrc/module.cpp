@@ -22,6 +22,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         &msbn::batch_norm_gather_stats_packed, py::arg("packed_all"),
         py::arg("running_mean"), py::arg("running_var"), py::arg("momentum"),
         py::arg("eps"));
+  m.def("batch_norm_gather_stats_packed_coefs",
+        &msbn::batch_norm_gather_stats_packed_coefs, py::arg("packed_all"),
+        py::arg("running_mean"), py::arg("running_var"), py::arg("momentum"),
+        py::arg("eps"), py::arg("weight"), py::arg("bias"),
+        py::arg("want_coefs"));
+  m.def("batch_norm_stats_local", &msbn::batch_norm_stats_local,
+        py::arg("input"), py::arg("eps"), py::arg("running_mean"),
+        py::arg("running_var"), py::arg("momentum"), py::arg("weight"),
+        py::arg("bias"), py::arg("want_coefs"));
   m.def("batch_norm_elemt", &msbn::batch_norm_elemt, py::arg("input"),
         py::arg("weight"), py::arg("bias"), py::arg("mean"), py::arg("invstd"),
         py::arg("eps"));
