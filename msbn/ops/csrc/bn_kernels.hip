@@ -138,11 +138,14 @@ __global__ void bn_stats_partial_nhwc(const T* __restrict__ x,
   }
 
   // LDS tree-reduce across the rowoff dimension (generic, non-pow2 safe).
+  // LDS layout [k][tid] with separate sum/sumsq planes: lane l of a wave
+  // lands on bank (2*l) mod 64 for ds_*_b64 -> conflict-free (the [tid][k]
+  // layout measured ~20 extra LDS cycles/instr via SQ_LDS_BANK_CONFLICT).
   __shared__ double sdata[MSBN_BLOCK * V * 2];
 #pragma unroll
   for (int k = 0; k < V; ++k) {
-    sdata[(threadIdx.x * V + k) * 2] = a[k];
-    sdata[(threadIdx.x * V + k) * 2 + 1] = b[k];
+    sdata[k * MSBN_BLOCK + threadIdx.x] = a[k];
+    sdata[(V + k) * MSBN_BLOCK + threadIdx.x] = b[k];
   }
   for (int st = 1; st < rpi; st <<= 1) {
     __syncthreads();
@@ -150,8 +153,9 @@ __global__ void bn_stats_partial_nhwc(const T* __restrict__ x,
       const int other = threadIdx.x + st * lpr;
 #pragma unroll
       for (int k = 0; k < V; ++k) {
-        sdata[(threadIdx.x * V + k) * 2] += sdata[(other * V + k) * 2];
-        sdata[(threadIdx.x * V + k) * 2 + 1] += sdata[(other * V + k) * 2 + 1];
+        sdata[k * MSBN_BLOCK + threadIdx.x] += sdata[k * MSBN_BLOCK + other];
+        sdata[(V + k) * MSBN_BLOCK + threadIdx.x] +=
+            sdata[(V + k) * MSBN_BLOCK + other];
       }
     }
   }
@@ -162,8 +166,8 @@ __global__ void bn_stats_partial_nhwc(const T* __restrict__ x,
     for (int k = 0; k < V; ++k) {
       if (c + k < C) {
         double* out = ws + ((c + k) * gridDim.y + chunk) * 2;
-        out[0] = sdata[(threadIdx.x * V + k) * 2];
-        out[1] = sdata[(threadIdx.x * V + k) * 2 + 1];
+        out[0] = sdata[k * MSBN_BLOCK + threadIdx.x];
+        out[1] = sdata[(V + k) * MSBN_BLOCK + threadIdx.x];
       }
     }
   }
@@ -556,11 +560,14 @@ __global__ void bn_bwd_reduce_partial_nhwc(const T* __restrict__ dy,
       }
     }
   }
+  // LDS layout [k][tid] with separate sum/sumsq planes: lane l of a wave
+  // lands on bank (2*l) mod 64 for ds_*_b64 -> conflict-free (the [tid][k]
+  // layout measured ~20 extra LDS cycles/instr via SQ_LDS_BANK_CONFLICT).
   __shared__ double sdata[MSBN_BLOCK * V * 2];
 #pragma unroll
   for (int k = 0; k < V; ++k) {
-    sdata[(threadIdx.x * V + k) * 2] = a[k];
-    sdata[(threadIdx.x * V + k) * 2 + 1] = b[k];
+    sdata[k * MSBN_BLOCK + threadIdx.x] = a[k];
+    sdata[(V + k) * MSBN_BLOCK + threadIdx.x] = b[k];
   }
   for (int st = 1; st < rpi; st <<= 1) {
     __syncthreads();
@@ -568,8 +575,9 @@ __global__ void bn_bwd_reduce_partial_nhwc(const T* __restrict__ dy,
       const int other = threadIdx.x + st * lpr;
 #pragma unroll
       for (int k = 0; k < V; ++k) {
-        sdata[(threadIdx.x * V + k) * 2] += sdata[(other * V + k) * 2];
-        sdata[(threadIdx.x * V + k) * 2 + 1] += sdata[(other * V + k) * 2 + 1];
+        sdata[k * MSBN_BLOCK + threadIdx.x] += sdata[k * MSBN_BLOCK + other];
+        sdata[(V + k) * MSBN_BLOCK + threadIdx.x] +=
+            sdata[(V + k) * MSBN_BLOCK + other];
       }
     }
   }
@@ -580,8 +588,8 @@ __global__ void bn_bwd_reduce_partial_nhwc(const T* __restrict__ dy,
     for (int k = 0; k < V; ++k) {
       if (c + k < C) {
         double* out = ws + ((c + k) * gridDim.y + chunk) * 2;
-        out[0] = sdata[(threadIdx.x * V + k) * 2];
-        out[1] = sdata[(threadIdx.x * V + k) * 2 + 1];
+        out[0] = sdata[k * MSBN_BLOCK + threadIdx.x];
+        out[1] = sdata[(V + k) * MSBN_BLOCK + threadIdx.x];
       }
     }
   }
