@@ -166,7 +166,18 @@ def _batch_norm_forward(module, input: Tensor, sync: bool) -> Tensor:
     )
 
     if not bn_training:
-        # Eval: normalize with running stats (differentiable wrt input/affine).
+        # Eval fast path: fused elemt kernel when no autograd is needed
+        # (inference serving); composed differentiable expression otherwise.
+        if input.is_cuda and not torch.is_grad_enabled():
+            from msbn import ops as _ops
+            from msbn.nn.functions import _contig
+
+            input = _contig(input)
+            rm = module.running_mean.to(torch.float32)
+            invstd = torch.rsqrt(module.running_var.to(torch.float32) + module.eps)
+            return _ops.batch_norm_elemt_act(
+                input, None, module.weight, module.bias, rm, invstd, False
+            )
         rm = module.running_mean.to(torch.float32)
         rv = module.running_var.to(torch.float32)
         invstd = torch.rsqrt(rv + module.eps)

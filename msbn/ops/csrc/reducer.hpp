@@ -236,6 +236,14 @@ class Reducer : public std::enable_shared_from_this<Reducer> {
     sync_enabled_ = enabled;
   }
 
+  // Gradient-compression hook support (stock register_comm_hook's builtin
+  // fp16/bf16 compress): buckets are cast to comm_dtype for the all-reduce
+  // and cast back before the grad copy-out.
+  void set_comm_dtype(c10::optional<at::ScalarType> dtype) {
+    std::lock_guard<std::mutex> lock(mutex_);
+    comm_dtype_ = dtype;
+  }
+
   // Called from the autograd post hook of parameter i.
   void autograd_hook(int64_t i) {
     std::lock_guard<std::mutex> lock(mutex_);
@@ -274,6 +282,10 @@ class Reducer : public std::enable_shared_from_this<Reducer> {
       if (b.work) {
         b.work->wait();
         b.work.reset();
+      }
+      if (b.wire.defined()) {
+        b.flat.copy_(b.wire, /*non_blocking=*/true);
+        b.wire = at::Tensor();
       }
       for (size_t k = 0; k < b.param_indices.size(); ++k) {
         auto& p = params_[b.param_indices[k]];
@@ -353,6 +365,7 @@ class Reducer : public std::enable_shared_from_this<Reducer> {
     bool ready = false;
     bool launched = false;
     c10::intrusive_ptr<c10d::Work> work;
+    at::Tensor wire;  // comm-dtype staging (gradient compression)
   };
 
   void initialize_buckets(std::vector<std::vector<int64_t>> bucket_indices) {
@@ -440,9 +453,17 @@ class Reducer : public std::enable_shared_from_this<Reducer> {
   void launch_bucket_locked(Bucket& b) {
     b.launched = true;
     if (div_factor_ != 1.0) b.flat.div_(div_factor_);
-    std::vector<at::Tensor> v{b.flat};
     c10d::AllreduceOptions opts;
-    b.work = pg_->allreduce(v, opts);
+    if (comm_dtype_.has_value() &&
+        b.flat.scalar_type() != *comm_dtype_) {
+      b.wire = b.flat.to(*comm_dtype_);
+      std::vector<at::Tensor> v{b.wire};
+      b.work = pg_->allreduce(v, opts);
+    } else {
+      b.wire = at::Tensor();
+      std::vector<at::Tensor> v{b.flat};
+      b.work = pg_->allreduce(v, opts);
+    }
   }
 
   std::vector<at::Tensor> params_;
@@ -457,6 +478,8 @@ class Reducer : public std::enable_shared_from_this<Reducer> {
   std::vector<bool> param_ready_;
   std::vector<std::shared_ptr<torch::autograd::Node>> grad_accumulators_;
   std::shared_ptr<bool> alive_ = std::make_shared<bool>(true);
+
+  c10::optional<at::ScalarType> comm_dtype_;
 
   std::mutex mutex_;
   bool expect_autograd_hooks_ = false;

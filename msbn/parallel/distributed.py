@@ -272,6 +272,27 @@ class DistributedDataParallel(Module):
         or drive the loop with run_with_join."""
         yield
 
+    def register_comm_hook(self, state, hook):
+        """Gradient-communication hook (stock register_comm_hook parity for
+        the builtin compression hooks): the fp16/bf16 compress hooks map to a
+        wire-dtype cast in the C++ reducer.  Arbitrary Python hooks are not
+        run inside the C++ backward path; use set_comm_dtype for custom
+        compression dtypes."""
+        name = getattr(hook, "__name__", repr(hook))
+        if "bf16" in name:
+            self.reducer.set_comm_dtype(torch.bfloat16)
+        elif "fp16" in name:
+            self.reducer.set_comm_dtype(torch.float16)
+        else:
+            raise NotImplementedError(
+                "msbn DDP supports the builtin fp16/bf16 compression hooks "
+                f"(got {name}); custom Python comm hooks are not supported"
+            )
+
+    def set_comm_dtype(self, dtype):
+        """Cast gradient buckets to `dtype` for the wire (None to disable)."""
+        self.reducer.set_comm_dtype(dtype)
+
     def _get_ddp_logging_data(self):
         return self.logger.data()
 

@@ -255,6 +255,25 @@ def test_checkpoint_roundtrip(tmp_path):
 
 
 # --------------------------------------------------------------------------
+def _comm_dtype_body(rank):
+    """bf16 wire compression: grads still averaged (to bf16 precision)."""
+    import msbn
+
+    torch.manual_seed(0)
+    net = msbn.parallel.DistributedDataParallel(torch.nn.Linear(8, 4))
+    net.set_comm_dtype(torch.bfloat16)
+    x = torch.full((2, 8), float(rank + 1))
+    net(x).sum().backward()
+    expect = (x.new_full((8,), 1.0 * 2) + x.new_full((8,), 2.0 * 2)) / 2
+    assert torch.allclose(net.module.weight.grad[0], expect, atol=0.05), (
+        net.module.weight.grad[0], expect)
+
+
+def test_comm_dtype_compression(tmp_path):
+    _spawn("_comm_dtype_body", tmp_path)
+
+
+# --------------------------------------------------------------------------
 def _join_uneven_body(rank):
     """run_with_join: rank 0 has 4 batches, rank 1 has 2; both must finish
     with identical (averaged) parameters and no hang."""
