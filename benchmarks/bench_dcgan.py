@@ -43,7 +43,9 @@ def main():
                                 rank=rank)
 
     torch.manual_seed(7)
-    G = msbn.convert_sync_batchnorm(msbn.models.Generator()).to(device)
+    from msbn.nn import fuse_bn_act
+
+    G = fuse_bn_act(msbn.convert_sync_batchnorm(msbn.models.Generator())).to(device)
     D = msbn.convert_sync_batchnorm(msbn.models.Discriminator()).to(device)
     if world > 1:
         G = msbn.parallel.DistributedDataParallel(

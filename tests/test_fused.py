@@ -98,3 +98,45 @@ def test_convert_leaves_fused_alone():
     n_after = sum(1 for x in out.modules()
                   if type(x).__name__ == "SyncBatchNormAct2d")
     assert n_before > 0 and n_after == n_before
+
+
+def test_fuse_bn_act_pass_dcgan():
+    """fuse_bn_act on DCGAN G: same function, BN+ReLU pairs become one module."""
+    import torch.nn as nn
+    from msbn.nn import fuse_bn_act
+
+    torch.manual_seed(3)
+    g1 = msbn.models.Generator(ngf=16)
+    g2 = msbn.models.Generator(ngf=16)
+    g2.load_state_dict(g1.state_dict())
+    g2 = fuse_bn_act(msbn.convert_sync_batchnorm(g2))
+    n_act = sum(1 for m in g2.modules()
+                if type(m).__name__ == "SyncBatchNormAct2d")
+    assert n_act == 4  # four BN+ReLU pairs in the DCGAN generator
+    assert not any(isinstance(m, nn.ReLU) for m in g2.modules())
+    g1.train(), g2.train()
+    z = torch.randn(3, 100, 1, 1)
+    y1, y2 = g1(z), g2(z)
+    torch.testing.assert_close(y1, y2, atol=1e-4, rtol=1e-4)
+    y1.sum().backward()
+    y2.sum().backward()
+    for (n1, p1), (n2, p2) in zip(g1.named_parameters(), g2.named_parameters()):
+        torch.testing.assert_close(p1.grad, p2.grad, atol=1e-3, rtol=1e-3), n1
+
+
+def test_fuse_bn_act_preserves_state():
+    import torch.nn as nn
+    from msbn.nn import fuse_bn_act
+
+    seq = nn.Sequential(
+        nn.Conv2d(3, 8, 3), nn.BatchNorm2d(8), nn.ReLU(),
+        nn.Conv2d(8, 8, 3), nn.BatchNorm2d(8),  # trailing BN without relu
+    )
+    with torch.no_grad():
+        seq[1].running_mean.fill_(0.25)
+    wrapped = nn.Sequential(seq)
+    out = fuse_bn_act(wrapped)
+    inner = out[0]
+    assert type(inner[1]).__name__ == "SyncBatchNormAct2d"
+    assert inner[1].running_mean[0].item() == 0.25
+    assert isinstance(inner[-1], nn.BatchNorm2d)  # untouched
