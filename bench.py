@@ -126,6 +126,7 @@ def main():
                 device_ids=[args.local_rank] if use_cuda else None,
                 output_device=args.local_rank if use_cuda else None,
                 gradient_as_bucket_view=True,
+                bucket_cap_mb=float(os.environ.get("MSBN_BUCKET_MB", "25")),
             )
     model.train()
 
@@ -148,7 +149,8 @@ def main():
         return loss
 
     graph = None
-    if args.graph and use_cuda and not distributed:
+    use_graph = args.graph or os.environ.get("MSBN_GRAPH", "0") == "1"
+    if use_graph and use_cuda and not distributed:
         # hipGraph-captured whole train step (fwd+bwd+optimizer): kills the
         # ~500 per-step kernel-launch round trips.  msbn's BN ops are
         # capture-safe by design (no host syncs; zero-count masking is
