@@ -89,6 +89,14 @@ class SyncBatchNormActFunction(torch.autograd.Function):
         need_bias_g = bias is not None and ctx.needs_input_grad[3]
 
         C = int(input.shape[1])
+        # Masked-grad materialization: when the residual branch needs its
+        # gradient, the reduce pass writes gm = dy*1[z>0] once; the elemt
+        # pass then reads gm instead of (dy, residual) and gm itself IS the
+        # residual gradient -> one fewer full-tensor read+write (8A -> 7A).
+        use_gm = (
+            relu and residual is not None and need_res_g and input.is_cuda
+            and input.numel() > 0
+        )
         if input.numel() == 0:
             # empty-input rank (join): contribute zeros, keep peers unblocked
             if world_size > 1 and (need_input_g or need_res_g):
@@ -104,10 +112,11 @@ class SyncBatchNormActFunction(torch.autograd.Function):
                 torch.zeros_like(bias) if need_bias_g else None,
                 None, None, None, None, None, None, None,
             )
+        gm = torch.empty_like(grad_output) if use_gm else None
         sum_dy, sum_dy_xmu, grad_weight, grad_bias = (
             ops.batch_norm_backward_reduce_act(
                 grad_output, input, residual, mean, invstd, weight, bias,
-                relu, need_input_g, need_weight_g, need_bias_g, coefs,
+                relu, need_input_g, need_weight_g, need_bias_g, coefs, gm,
             )
         )
         grad_input = grad_res = None
@@ -119,10 +128,17 @@ class SyncBatchNormActFunction(torch.autograd.Function):
                 comm_log.record("all_reduce", combined.numel() * 4,
                                 f"syncbn bwd C={C}")
                 sum_dy, sum_dy_xmu = combined[:C], combined[C:]
-            grad_input, grad_res = ops.batch_norm_backward_elemt_act(
-                grad_output, input, residual, mean, invstd, weight, bias,
-                sum_dy, sum_dy_xmu, count_sum, relu, need_res_g, coefs,
-            )
+            if use_gm:
+                grad_input, _ = ops.batch_norm_backward_elemt_act(
+                    gm, input, None, mean, invstd, weight, bias,
+                    sum_dy, sum_dy_xmu, count_sum, False, False, coefs,
+                )
+                grad_res = gm
+            else:
+                grad_input, grad_res = ops.batch_norm_backward_elemt_act(
+                    grad_output, input, residual, mean, invstd, weight, bias,
+                    sum_dy, sum_dy_xmu, count_sum, relu, need_res_g, coefs,
+                )
         return (
             grad_input if need_input_g else None,
             grad_res,

@@ -186,15 +186,15 @@ def batch_norm_elemt_act(input, residual, weight, bias, mean, invstd,
 
 def batch_norm_backward_reduce_act(grad_out, input, residual, mean, invstd,
                                    weight, bias, relu_mask, input_g, weight_g,
-                                   bias_g, coefs=None):
+                                   bias_g, coefs=None, gm_out=None):
     if input.is_cuda:
         return _require_hip().batch_norm_backward_reduce_act(
             grad_out, input, residual, mean, invstd, weight, bias, relu_mask,
-            input_g, weight_g, bias_g, coefs,
+            input_g, weight_g, bias_g, coefs, gm_out,
         )
     return _ref.batch_norm_backward_reduce_act(
         grad_out, input, residual, mean, invstd, weight, bias, relu_mask,
-        input_g, weight_g, bias_g,
+        input_g, weight_g, bias_g, gm_out,
     )
 
 

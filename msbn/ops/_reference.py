@@ -170,10 +170,12 @@ def _masked_grad(grad_out, input, residual, mean, invstd, weight, bias,
 
 def batch_norm_backward_reduce_act(
     grad_out, input, residual, mean, invstd, weight, bias, relu_mask,
-    input_g, weight_g, bias_g,
+    input_g, weight_g, bias_g, gm_out=None,
 ):
     g = _masked_grad(grad_out, input, residual, mean, invstd, weight, bias,
                      relu_mask)
+    if gm_out is not None:
+        gm_out.copy_(g.to(grad_out.dtype))
     return batch_norm_backward_reduce(
         g, input, mean, invstd, weight, input_g, weight_g, bias_g
     )

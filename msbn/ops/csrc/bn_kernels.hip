@@ -404,10 +404,11 @@ __global__ void bn_elemt_nhwc(const T* __restrict__ x,
 // z = scale[c]*x + shift[c] (+ res) is recomputed in-kernel, so the fused
 // forward never has to store a mask or the pre-activation tensor.
 // =====================================================================
-template <typename T, int V, bool MASK, bool RES>
+template <typename T, int V, bool MASK, bool RES, bool GMOUT>
 __global__ void bn_bwd_reduce_partial_nchw(const T* __restrict__ dy,
                                            const T* __restrict__ x,
                                            const T* __restrict__ res,
+                                           T* __restrict__ gm_out,
                                            const float* __restrict__ mean,
                                            const float* __restrict__ scale,
                                            const float* __restrict__ shift,
@@ -434,6 +435,7 @@ __global__ void bn_bwd_reduce_partial_nchw(const T* __restrict__ dy,
           if (RES) z += to_f(res[base + s]);
           if (z <= 0.f) g = 0.f;
         }
+        if (GMOUT) gm_out[base + s] = from_f<T>(g);
         a += g;
         b += (double)g * (xv - m);
       }
@@ -442,7 +444,7 @@ __global__ void bn_bwd_reduce_partial_nchw(const T* __restrict__ dy,
            s += (int64_t)blockDim.x * V) {
         Pack<T, V> pg = *reinterpret_cast<const Pack<T, V>*>(&dy[base + s]);
         Pack<T, V> px = *reinterpret_cast<const Pack<T, V>*>(&x[base + s]);
-        Pack<T, V> pr;
+        Pack<T, V> pr, pm;
         if (RES) pr = *reinterpret_cast<const Pack<T, V>*>(&res[base + s]);
 #pragma unroll
         for (int k = 0; k < V; ++k) {
@@ -453,9 +455,11 @@ __global__ void bn_bwd_reduce_partial_nchw(const T* __restrict__ dy,
             if (RES) z += to_f(pr.v[k]);
             if (z <= 0.f) g = 0.f;
           }
+          if (GMOUT) pm.v[k] = from_f<T>(g);
           a += g;
           b += (double)g * (xv - m);
         }
+        if (GMOUT) *reinterpret_cast<Pack<T, V>*>(&gm_out[base + s]) = pm;
       }
     }
   }
@@ -469,10 +473,11 @@ __global__ void bn_bwd_reduce_partial_nchw(const T* __restrict__ dy,
   }
 }
 
-template <typename T, bool MASK, bool RES>
+template <typename T, bool MASK, bool RES, bool GMOUT>
 __global__ void bn_bwd_reduce_partial_nchw_flat(
     const T* __restrict__ dy, const T* __restrict__ x,
-    const T* __restrict__ res, const float* __restrict__ mean,
+    const T* __restrict__ res, T* __restrict__ gm_out,
+    const float* __restrict__ mean,
     const float* __restrict__ scale, const float* __restrict__ shift,
     double* __restrict__ ws, int64_t N, int64_t C, int64_t S,
     int64_t chunk_len) {
@@ -494,6 +499,7 @@ __global__ void bn_bwd_reduce_partial_nchw_flat(
       if (RES) z += to_f(res[e]);
       if (z <= 0.f) g = 0.f;
     }
+    if (GMOUT) gm_out[e] = from_f<T>(g);
     a += g;
     b += (double)g * (xv - m);
   }
@@ -506,10 +512,11 @@ __global__ void bn_bwd_reduce_partial_nchw_flat(
   }
 }
 
-template <typename T, int V, bool MASK, bool RES>
+template <typename T, int V, bool MASK, bool RES, bool GMOUT>
 __global__ void bn_bwd_reduce_partial_nhwc(const T* __restrict__ dy,
                                            const T* __restrict__ x,
                                            const T* __restrict__ res,
+                                           T* __restrict__ gm_out,
                                            const float* __restrict__ mean,
                                            const float* __restrict__ scale,
                                            const float* __restrict__ shift,
@@ -544,7 +551,7 @@ __global__ void bn_bwd_reduce_partial_nhwc(const T* __restrict__ dy,
     for (int64_t r = r0 + rowoff; r < r1; r += rpi) {
       Pack<T, V> pg = *reinterpret_cast<const Pack<T, V>*>(&dy[r * C + c]);
       Pack<T, V> px = *reinterpret_cast<const Pack<T, V>*>(&x[r * C + c]);
-      Pack<T, V> pr;
+      Pack<T, V> pr, pm;
       if (RES) pr = *reinterpret_cast<const Pack<T, V>*>(&res[r * C + c]);
 #pragma unroll
       for (int k = 0; k < V; ++k) {
@@ -555,9 +562,11 @@ __global__ void bn_bwd_reduce_partial_nhwc(const T* __restrict__ dy,
           if (RES) z += to_f(pr.v[k]);
           if (z <= 0.f) g = 0.f;
         }
+        if (GMOUT) pm.v[k] = from_f<T>(g);
         a[k] += g;
         b[k] += (double)g * (xv - m[k]);
       }
+      if (GMOUT) *reinterpret_cast<Pack<T, V>*>(&gm_out[r * C + c]) = pm;
     }
   }
   // LDS layout [k][tid] with separate sum/sumsq planes: lane l of a wave
@@ -1316,8 +1325,16 @@ batch_norm_backward_reduce_act(
     const c10::optional<at::Tensor>& residual, const at::Tensor& mean,
     const at::Tensor& invstd, const c10::optional<at::Tensor>& weight,
     const c10::optional<at::Tensor>& bias, bool relu_mask, bool input_g,
-    bool weight_g, bool bias_g, const c10::optional<at::Tensor>& coefs_in) {
+    bool weight_g, bool bias_g, const c10::optional<at::Tensor>& coefs_in,
+    const c10::optional<at::Tensor>& gm_out) {
   const Layout L = get_layout(input);
+  const bool want_gm = gm_out.has_value();
+  if (want_gm) {
+    TORCH_CHECK(gm_out->sizes() == input.sizes() &&
+                    gm_out->strides() == input.strides() &&
+                    gm_out->scalar_type() == input.scalar_type(),
+                "gm_out must match input shape/strides/dtype");
+  }
   TORCH_CHECK(grad_out.sizes() == input.sizes() &&
                   grad_out.strides() == input.strides(),
               "grad_out must match input shape and layout");
@@ -1358,11 +1375,14 @@ batch_norm_backward_reduce_act(
     const native_t* res =
         has_res ? reinterpret_cast<const native_t*>(residual->data_ptr())
                 : nullptr;
+    native_t* gm =
+        want_gm ? reinterpret_cast<native_t*>(gm_out->data_ptr()) : nullptr;
     constexpr int VMAX = 16 / (int)sizeof(native_t);
     at::Tensor ws;
     int nchunks = 0;
     MSBN_DISPATCH_BOOL(relu_mask, MASK_, [&] {
       MSBN_DISPATCH_BOOL(has_res, RES_, [&] {
+        MSBN_DISPATCH_BOOL(want_gm, GM_, [&] {
         if (!L.nhwc) {
           const int v = pick_v<native_t>(x, dy, res, L.S);
           if (v == 1) {
@@ -1371,8 +1391,8 @@ batch_norm_backward_reduce_act(
             ws = at::empty({(int64_t)nchunks * L.C * 2},
                            input.options().dtype(at::kDouble));
             hipLaunchKernelGGL(
-                (bn_bwd_reduce_partial_nchw_flat<native_t, MASK_, RES_>),
-                g.grid, dim3(MSBN_BLOCK), 0, stream, dy, x, res,
+                (bn_bwd_reduce_partial_nchw_flat<native_t, MASK_, RES_, GM_>),
+                g.grid, dim3(MSBN_BLOCK), 0, stream, dy, x, res, gm,
                 mean.data_ptr<float>(), scale, shift, ws.data_ptr<double>(),
                 L.N, L.C, L.S, g.chunk_len);
           } else {
@@ -1382,8 +1402,8 @@ batch_norm_backward_reduce_act(
                            input.options().dtype(at::kDouble));
             MSBN_DISPATCH_V(v, VMAX, [&] {
               hipLaunchKernelGGL(
-                  (bn_bwd_reduce_partial_nchw<native_t, VV, MASK_, RES_>),
-                  g.grid, dim3(MSBN_BLOCK), 0, stream, dy, x, res,
+                  (bn_bwd_reduce_partial_nchw<native_t, VV, MASK_, RES_, GM_>),
+                  g.grid, dim3(MSBN_BLOCK), 0, stream, dy, x, res, gm,
                   mean.data_ptr<float>(), scale, shift, ws.data_ptr<double>(),
                   L.N, L.C, L.S, g.chunkN, g.chunkS);
             });
@@ -1396,12 +1416,13 @@ batch_norm_backward_reduce_act(
                          input.options().dtype(at::kDouble));
           MSBN_DISPATCH_V(v, VMAX, [&] {
             hipLaunchKernelGGL(
-                (bn_bwd_reduce_partial_nhwc<native_t, VV, MASK_, RES_>),
-                g.grid, dim3(MSBN_BLOCK), 0, stream, dy, x, res,
+                (bn_bwd_reduce_partial_nhwc<native_t, VV, MASK_, RES_, GM_>),
+                g.grid, dim3(MSBN_BLOCK), 0, stream, dy, x, res, gm,
                 mean.data_ptr<float>(), scale, shift, ws.data_ptr<double>(),
                 L.rows, L.C, g.chunk_rows, g.lpr);
           });
         }
+        });
       });
     });
     const int fgrid = (int)cdiv(L.C, kFinalizeWavesPerBlock);
@@ -1429,7 +1450,7 @@ batch_norm_backward_reduce(const at::Tensor& grad_out, const at::Tensor& input,
   return batch_norm_backward_reduce_act(grad_out, input, c10::nullopt, mean,
                                         invstd, weight, c10::nullopt, false,
                                         input_g, weight_g, bias_g,
-                                        c10::nullopt);
+                                        c10::nullopt, c10::nullopt);
 }
 
 std::tuple<at::Tensor, at::Tensor> batch_norm_backward_elemt_act(

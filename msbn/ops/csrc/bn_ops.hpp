@@ -98,7 +98,8 @@ batch_norm_backward_reduce_act(
     const c10::optional<at::Tensor>& residual, const at::Tensor& mean,
     const at::Tensor& invstd, const c10::optional<at::Tensor>& weight,
     const c10::optional<at::Tensor>& bias, bool relu_mask, bool input_g,
-    bool weight_g, bool bias_g, const c10::optional<at::Tensor>& coefs_in);
+    bool weight_g, bool bias_g, const c10::optional<at::Tensor>& coefs_in,
+    const c10::optional<at::Tensor>& gm_out);
 
 // returns (grad_input, grad_residual-or-undefined)
 std::tuple<at::Tensor, at::Tensor> batch_norm_backward_elemt_act(

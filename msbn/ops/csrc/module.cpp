@@ -53,7 +53,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("input"), py::arg("residual"), py::arg("mean"),
         py::arg("invstd"), py::arg("weight"), py::arg("bias"),
         py::arg("relu_mask"), py::arg("input_g"), py::arg("weight_g"),
-        py::arg("bias_g"), py::arg("coefs") = py::none());
+        py::arg("bias_g"), py::arg("coefs") = py::none(),
+        py::arg("gm_out") = py::none());
   m.def("batch_norm_backward_elemt_act", &msbn::batch_norm_backward_elemt_act,
         py::arg("grad_out"), py::arg("input"), py::arg("residual"),
         py::arg("mean"), py::arg("invstd"), py::arg("weight"), py::arg("bias"),
