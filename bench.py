@@ -130,8 +130,12 @@ def main():
             )
     model.train()
 
-    opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9,
-                          weight_decay=1e-4)
+    try:  # fused foreach SGD (works on ROCm; identical for both impls)
+        opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9,
+                              weight_decay=1e-4, fused=True)
+    except (RuntimeError, TypeError):
+        opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9,
+                              weight_decay=1e-4)
     loss_fn = torch.nn.CrossEntropyLoss()
 
     bs = args.batch_size
