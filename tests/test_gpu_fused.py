@@ -224,3 +224,35 @@ def test_fused_equals_unfused_resnet_gpu():
         cos = torch.nn.functional.cosine_similarity(ga, gb, dim=0).item()
         rel = (ga - gb).norm().item() / (ga.norm().item() + 1e-12)
         assert cos > 0.999 and rel < 0.05, (n1, cos, rel)
+
+
+def test_graphed_step_gpu():
+    """hipGraph capture of a fused SyncBN model train step; replays update
+    params and running stats."""
+    from msbn.utils import GraphedStep
+
+    torch.manual_seed(21)
+    model = msbn.models.resnet18(fused=True).to(DEV)
+    model.train()
+    opt = torch.optim.SGD(model.parameters(), lr=0.01, momentum=0.9)
+    x = torch.randn(8, 3, 64, 64, device=DEV)
+    y = torch.randint(0, 1000, (8,), device=DEV)
+
+    def step():
+        opt.zero_grad(set_to_none=False)
+        loss = torch.nn.functional.cross_entropy(model(x), y)
+        loss.backward()
+        opt.step()
+        return loss
+
+    nbt0 = model.bn1.num_batches_tracked.item()
+    g = GraphedStep(step, warmup=3)
+    w0 = model.fc.weight.detach().clone()
+    rm0 = model.bn1.running_mean.detach().clone()
+    for _ in range(3):
+        g.replay()
+    torch.cuda.synchronize()
+    assert not torch.equal(model.fc.weight, w0), "params must move on replay"
+    assert not torch.equal(model.bn1.running_mean, rm0), \
+        "running stats must update on replay"
+    assert model.bn1.num_batches_tracked.item() > nbt0
