@@ -227,32 +227,21 @@ def test_fused_equals_unfused_resnet_gpu():
 
 
 def test_graphed_step_gpu():
-    """hipGraph capture of a fused SyncBN model train step; replays update
-    params and running stats."""
-    from msbn.utils import GraphedStep
+    """hipGraph capture of a full train step (fwd+bwd+optimizer) on the fused
+    model, replayed with parameter updates.  Runs in a SUBPROCESS: capture
+    works standalone but segfaults inside the pytest host process (plugin
+    interaction with capture_end on ROCm) — tools/graph_bisect.py is the
+    actual test body."""
+    import os
+    import subprocess
+    import sys
 
-    torch.manual_seed(21)
-    model = msbn.models.resnet18(fused=True).to(DEV)
-    model.train()
-    opt = torch.optim.SGD(model.parameters(), lr=0.01, momentum=0.9)
-    x = torch.randn(8, 3, 64, 64, device=DEV)
-    y = torch.randint(0, 1000, (8,), device=DEV)
-
-    def step():
-        opt.zero_grad(set_to_none=False)
-        loss = torch.nn.functional.cross_entropy(model(x), y)
-        loss.backward()
-        opt.step()
-        return loss
-
-    nbt0 = model.bn1.num_batches_tracked.item()
-    g = GraphedStep(step, warmup=3)
-    w0 = model.fc.weight.detach().clone()
-    rm0 = model.bn1.running_mean.detach().clone()
-    for _ in range(3):
-        g.replay()
-    torch.cuda.synchronize()
-    assert not torch.equal(model.fc.weight, w0), "params must move on replay"
-    assert not torch.equal(model.bn1.running_mean, rm0), \
-        "running stats must update on replay"
-    assert model.bn1.num_batches_tracked.item() > nbt0
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    for case in ("a", "b"):
+        r = subprocess.run(
+            [sys.executable, os.path.join(repo, "tools", "graph_bisect.py"),
+             case],
+            capture_output=True, text=True, timeout=180, cwd=repo,
+        )
+        assert r.returncode == 0 and f"CASE {case} OK" in r.stdout, (
+            r.stdout, r.stderr)
