@@ -86,6 +86,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
            py::arg("unused_params") = std::vector<int64_t>())
       .def("set_grad_sync_enabled", &msbn::Reducer::set_grad_sync_enabled)
       .def("set_comm_dtype", &msbn::Reducer::set_comm_dtype)
+      .def("set_nan_check", &msbn::Reducer::set_nan_check)
       .def("finalize_backward", &msbn::Reducer::finalize_backward,
            py::call_guard<py::gil_scoped_release>())
       .def("rebuild_buckets", &msbn::Reducer::rebuild_buckets,

@@ -244,6 +244,14 @@ class Reducer : public std::enable_shared_from_this<Reducer> {
     comm_dtype_ = dtype;
   }
 
+  // Debug guard (stock TORCH_NCCL_NAN_CHECK analog): raise before shipping a
+  // bucket containing non-finite gradients.  Costs a device reduction +
+  // host sync per bucket — debugging only.
+  void set_nan_check(bool enabled) {
+    std::lock_guard<std::mutex> lock(mutex_);
+    nan_check_ = enabled;
+  }
+
   // Called from the autograd post hook of parameter i.
   void autograd_hook(int64_t i) {
     std::lock_guard<std::mutex> lock(mutex_);
@@ -452,6 +460,11 @@ class Reducer : public std::enable_shared_from_this<Reducer> {
 
   void launch_bucket_locked(Bucket& b) {
     b.launched = true;
+    if (nan_check_) {
+      TORCH_CHECK(at::isfinite(b.flat).all().item<bool>(),
+                  "msbn Reducer: non-finite gradient detected in bucket "
+                  "(params ", b.param_indices, ") before all-reduce");
+    }
     if (div_factor_ != 1.0) b.flat.div_(div_factor_);
     c10d::AllreduceOptions opts;
     if (comm_dtype_.has_value() &&
@@ -480,6 +493,7 @@ class Reducer : public std::enable_shared_from_this<Reducer> {
   std::shared_ptr<bool> alive_ = std::make_shared<bool>(true);
 
   c10::optional<at::ScalarType> comm_dtype_;
+  bool nan_check_ = false;
 
   std::mutex mutex_;
   bool expect_autograd_hooks_ = false;

@@ -143,6 +143,10 @@ class DistributedDataParallel(Module):
             _DEFAULT_FIRST_BUCKET_BYTES,
             self.bucket_bytes_cap,
         )
+        import os as _os
+
+        if _os.environ.get("MSBN_NAN_CHECK", "0") == "1":
+            self.reducer.set_nan_check(True)
         self.logger = DDPLogger(self)
         self._has_sync_bn = any(
             type(m).__name__ == "SyncBatchNorm" for m in module.modules()

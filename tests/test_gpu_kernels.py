@@ -234,3 +234,49 @@ def test_resnet18_bf16_step_gpu():
     opt.step()
     torch.cuda.synchronize()
     assert torch.isfinite(loss)
+
+
+def test_batchnorm3d_module_gpu():
+    """BatchNorm3d / SyncBatchNorm on 5-D input incl. channels_last_3d."""
+    torch.manual_seed(31)
+    C = 12
+    ours = msbn.nn.SyncBatchNorm(C).to(DEV)
+    theirs = torch.nn.BatchNorm3d(C).to(DEV)
+    with torch.no_grad():
+        theirs.weight.copy_(ours.weight)
+        theirs.bias.copy_(ours.bias)
+    x = torch.randn(3, C, 4, 6, 5, device=DEV)
+    for fmt in (torch.contiguous_format, torch.channels_last_3d):
+        x1 = x.to(memory_format=fmt).requires_grad_(True)
+        x2 = x.clone().requires_grad_(True)
+        y1 = ours(x1)
+        y2 = theirs(x2)
+        torch.testing.assert_close(
+            y1.contiguous(), y2, atol=1e-4, rtol=1e-4
+        )
+        y1.sum().backward()
+        y2.sum().backward()
+        torch.testing.assert_close(
+            x1.grad.contiguous(), x2.grad, atol=1e-4, rtol=1e-4
+        )
+        ours.zero_grad(set_to_none=True)
+        theirs.zero_grad(set_to_none=True)
+
+
+def test_batchnorm1d_module_gpu():
+    torch.manual_seed(32)
+    C = 20
+    ours = msbn.nn.BatchNorm1d(C).to(DEV)
+    theirs = torch.nn.BatchNorm1d(C).to(DEV)
+    with torch.no_grad():
+        theirs.weight.copy_(ours.weight)
+        theirs.bias.copy_(ours.bias)
+    for shape in ((8, C), (4, C, 11)):
+        x = torch.randn(*shape, device=DEV)
+        x1 = x.clone().requires_grad_(True)
+        x2 = x.clone().requires_grad_(True)
+        y1, y2 = ours(x1), theirs(x2)
+        torch.testing.assert_close(y1, y2, atol=1e-4, rtol=1e-4)
+        (y1.pow(2).sum()).backward()
+        (y2.pow(2).sum()).backward()
+        torch.testing.assert_close(x1.grad, x2.grad, atol=1e-3, rtol=1e-3)

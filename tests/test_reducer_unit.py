@@ -120,3 +120,21 @@ def test_reducer_rebuild_world1(tmp_path):
     red.prepare_for_backward([])
     m(torch.randn(4, 8)).sum().backward()
     assert all(p.grad is not None for p in params)
+
+
+def test_reducer_nan_check(tmp_path):
+    _init_world1(tmp_path)
+    m = torch.nn.Linear(4, 4)
+    params = list(m.parameters())
+    red = C.Reducer(params, [[0, 1]], dist.group.WORLD, False, 1 << 20,
+                    25 << 20)
+    red.set_nan_check(True)
+    red.prepare_for_backward([])
+    x = torch.full((2, 4), float("inf"))
+    with pytest.raises(RuntimeError, match="non-finite"):
+        (m(x) * 0 + m(x)).sum().backward()
+    # recover: clear the poisoned grads, reducer usable again
+    m.zero_grad(set_to_none=True)
+    red.prepare_for_backward([])
+    m(torch.randn(2, 4)).sum().backward()
+    assert red.iterations() >= 1
