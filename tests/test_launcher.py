@@ -94,3 +94,20 @@ def test_restart_on_failure(tmp_path):
     )
     assert r.returncode == 0, r.stderr
     assert "SECOND_TRY_OK" in r.stdout
+
+
+def test_example_script_end_to_end(tmp_path):
+    """The shipped reference-recipe example runs under msbn.launch (CPU/gloo)."""
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO
+    ck = tmp_path / "ex.pt"
+    r = subprocess.run(
+        [sys.executable, "-m", "msbn.launch", "--nproc_per_node=2",
+         os.path.join(REPO, "examples", "distributed_train.py"),
+         "--ngpu", "2", "--epochs", "1", "--batch-size", "8",
+         "--ckpt", str(ck)],
+        capture_output=True, text=True, timeout=300, env=env, cwd=REPO,
+    )
+    assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
+    assert "TRAIN_OK" in r.stdout
+    assert ck.exists()
