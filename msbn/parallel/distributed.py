@@ -79,8 +79,7 @@ class DistributedDataParallel(Module):
         static_graph: bool = False,
     ):
         super().__init__()
-        from msbn import ops as _ops
-        from msbn import _C as _C  # noqa: F401  (reducer lives here)
+        from msbn import _C as _C  # noqa: F401  fail fast if ext missing
         if not dist.is_available() or not dist.is_initialized():
             raise RuntimeError(
                 "msbn DDP requires torch.distributed to be initialized "
@@ -119,7 +118,6 @@ class DistributedDataParallel(Module):
                     "(README.md:51-53)"
                 )
         self._params = params
-        self._param_index = {id(p): i for i, p in enumerate(params)}
 
         import msbn._C as C
 

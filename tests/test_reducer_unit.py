@@ -1,13 +1,11 @@
 """Unit tests of the C++ reducer machinery that need no process group >1:
 bucket assignment + single-process (world_size 1, gloo) reducer behavior."""
 
-import os
-
 import pytest
 import torch
 import torch.distributed as dist
 
-import msbn
+import msbn  # noqa: F401  (registers the extension)
 import msbn._C as C
 
 

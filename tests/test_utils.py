@@ -1,7 +1,6 @@
 """utils coverage: logger fields, master_print, env helpers, comm log."""
 
 import io
-import os
 
 import torch
 
