@@ -48,6 +48,7 @@ def one_case(rng):
     relu = rng.random() < 0.5
     with_res = rng.random() < 0.5
     desc = f"shape={shape} dtype={dtype} cl={cl} relu={relu} res={with_res}"
+    one_case.last = desc
 
     fmt = (torch.channels_last if dims == 4 else torch.channels_last_3d) \
         if cl else torch.contiguous_format
@@ -119,7 +120,8 @@ def main():
         try:
             one_case(rng)
         except AssertionError as e:
-            print(f"FUZZ FAIL after {cases} cases:\n{e}")
+            print(f"FUZZ FAIL after {cases} cases: "
+                  f"{getattr(one_case, 'last', '?')}\n{e}")
             sys.exit(1)
         cases += 1
     print(f"FUZZ OK: {cases} cases in {time.time() - t0:.0f}s")
