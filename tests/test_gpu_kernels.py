@@ -247,8 +247,8 @@ def test_batchnorm3d_module_gpu():
         theirs.bias.copy_(ours.bias)
     x = torch.randn(3, C, 4, 6, 5, device=DEV)
     for fmt in (torch.contiguous_format, torch.channels_last_3d):
-        x1 = x.to(memory_format=fmt).requires_grad_(True)
-        x2 = x.clone().requires_grad_(True)
+        x1 = x.detach().clone().to(memory_format=fmt).requires_grad_(True)
+        x2 = x.detach().clone().requires_grad_(True)
         y1 = ours(x1)
         y2 = theirs(x2)
         torch.testing.assert_close(

@@ -99,11 +99,23 @@ def one_case(rng):
         g.float(), x.float(), None if res is None else res.float(),
         mean_r, invstd_r, w, b, rsdy, rsdyx, torch.tensor([float(n)]),
         relu, with_res)
+    # exclude ReLU-boundary elements (|z| ~ 0): a 1-ulp stats difference
+    # between implementations legitimately flips the gate there
+    if relu:
+        scale_r, shift_r = ref._act_coefs(mean_r, invstd_r, w, b)
+        z = x.float() * ref._chan_view(scale_r, x) + ref._chan_view(shift_r, x)
+        if res is not None:
+            z = z + res.float()
+        zthr = 1e-5 if dtype == torch.float32 else 2e-2
+        interior = (z.abs() > zthr)
+    else:
+        interior = torch.ones_like(x, dtype=torch.bool)
     istd_scale = float(invstd_r.max().clamp(min=1.0))
-    torch.testing.assert_close(dx.float().cpu(), rdx,
+    torch.testing.assert_close(dx.float().cpu()[interior], rdx[interior],
                                **tol(dtype, 10 * istd_scale))
     if with_res:
-        torch.testing.assert_close(dres.float().cpu(), rdres, **tol(dtype, 2))
+        torch.testing.assert_close(dres.float().cpu()[interior],
+                                   rdres[interior], **tol(dtype, 2))
     return desc
 
 
