@@ -1328,6 +1328,17 @@ batch_norm_backward_reduce_act(
     bool weight_g, bool bias_g, const c10::optional<at::Tensor>& coefs_in,
     const c10::optional<at::Tensor>& gm_out) {
   const Layout L = get_layout(input);
+  if (input.numel() == 0) {
+    auto z32 = input.options().dtype(at::kFloat);
+    auto combined0 = at::zeros({2 * L.C}, z32);
+    const auto wt = weight.has_value() ? weight->scalar_type()
+                                       : input.scalar_type();
+    at::Tensor gw0, gb0;
+    if (weight_g) gw0 = at::zeros({L.C}, input.options().dtype(wt));
+    if (bias_g) gb0 = at::zeros({L.C}, input.options().dtype(wt));
+    return {combined0.narrow(0, 0, L.C), combined0.narrow(0, L.C, L.C), gw0,
+            gb0};
+  }
   const bool want_gm = gm_out.has_value();
   if (want_gm) {
     TORCH_CHECK(gm_out->sizes() == input.sizes() &&

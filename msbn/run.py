@@ -57,6 +57,15 @@ def parse_args(argv: Optional[List[str]] = None, use_env_default: bool = True):
     )
     p.add_argument("--master-port", "--master_port", type=int, default=None)
     p.add_argument("--max-restarts", "--max_restarts", type=int, default=0)
+    p.add_argument("--standalone", action="store_true",
+                   help="single-node shortcut (torchrun compat): implies "
+                        "127.0.0.1 rendezvous on a free port")
+    p.add_argument("--local-addr", "--local_addr", type=str, default=None,
+                   help="accepted for torchrun compatibility")
+    p.add_argument("--rdzv-backend", "--rdzv_backend", type=str, default=None,
+                   help="accepted for torchrun compatibility (static only)")
+    p.add_argument("--rdzv-endpoint", "--rdzv_endpoint", type=str, default=None,
+                   help="host:port (torchrun compat); overrides master addr/port")
     p.add_argument(
         "--use-env", "--use_env", action="store_true", default=use_env_default,
         help="pass LOCAL_RANK via env only (no --local-rank argv)",
@@ -125,6 +134,15 @@ def _kill_group(procs):
 
 
 def run(args) -> int:
+    if getattr(args, "rdzv_endpoint", None):
+        host, _, port = args.rdzv_endpoint.partition(":")
+        if host:
+            args.master_addr = host
+        if port:
+            args.master_port = int(port)
+    if getattr(args, "standalone", False):
+        args.master_addr = "127.0.0.1"
+        args.master_port = None
     nproc = (
         _device_count()
         if str(args.nproc_per_node) in ("auto", "gpu")

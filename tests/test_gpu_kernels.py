@@ -280,3 +280,17 @@ def test_batchnorm1d_module_gpu():
         (y1.pow(2).sum()).backward()
         (y2.pow(2).sum()).backward()
         torch.testing.assert_close(x1.grad, x2.grad, atol=1e-3, rtol=1e-3)
+
+
+def test_backward_reduce_empty_input_gpu():
+    """Empty input through the public backward-reduce op (host-side grid
+    guard; callers normally gate on count)."""
+    x = torch.empty(0, 8, 4, 4, device=DEV)
+    g = torch.empty(0, 8, 4, 4, device=DEV)
+    mean = torch.zeros(8, device=DEV)
+    invstd = torch.zeros(8, device=DEV)
+    sdy, sdyx, gw, gb = ops.batch_norm_backward_reduce(
+        g, x, mean, invstd, None, True, True, True
+    )
+    assert torch.all(sdy == 0) and torch.all(sdyx == 0)
+    assert torch.all(gw == 0) and torch.all(gb == 0)
