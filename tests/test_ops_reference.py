@@ -163,3 +163,34 @@ def test_2d_input():
     assert torch.allclose(y, ref, atol=1e-5)
     y.sum().backward()
     assert x.grad is not None
+
+
+def test_no_affine_and_no_tracking():
+    """affine=False and track_running_stats=False module paths."""
+    torch.manual_seed(8)
+    bn = msbn.nn.SyncBatchNorm(6, affine=False, track_running_stats=False)
+    assert bn.weight is None and bn.running_mean is None
+    x = torch.randn(4, 6, 5, requires_grad=True)
+    bn.train()
+    y = bn(x)
+    ref = F.batch_norm(x, None, None, None, None, training=True, eps=bn.eps)
+    assert torch.allclose(y, ref, atol=1e-5)
+    y.sum().backward()
+    assert x.grad is not None
+    # eval without running stats keeps using batch stats (bn_training True)
+    bn.eval()
+    y2 = bn(torch.randn(4, 6, 5))
+    assert y2.shape == (4, 6, 5)
+
+
+def test_momentum_none_cumulative():
+    """momentum=None -> cumulative moving average (stock batchnorm.py:754-765)."""
+    torch.manual_seed(9)
+    ours = msbn.nn.SyncBatchNorm(3, momentum=None)
+    theirs = torch.nn.BatchNorm1d(3, momentum=None)
+    for _ in range(5):
+        x = torch.randn(6, 3)
+        ours.train()(x)
+        theirs.train()(x)
+    assert torch.allclose(ours.running_mean, theirs.running_mean, atol=1e-6)
+    assert torch.allclose(ours.running_var, theirs.running_var, atol=1e-5)
