@@ -68,19 +68,15 @@ def main():
     if use_cuda:
         x = x.to(memory_format=torch.channels_last)
 
-    inner = model.module if hasattr(model, "module") else model
-
     def step():
         opt.zero_grad(set_to_none=True)
-        if world > 1:
-            # run through the DDP wrapper for bucket hooks
-            cls_outs, box_outs = model(x)
-            loss = x.new_zeros((), dtype=torch.float32)
-            for c, b in zip(cls_outs, box_outs):
-                loss = loss + torch.sigmoid(c.float()).pow(2).mean() \
-                    + b.float().pow(2).mean()
-        else:
-            loss = inner.training_loss(x)
+        cls_outs, box_outs = model(x)
+        # dense synthetic objective, entirely on-device (identical for the
+        # single- and multi-GPU paths)
+        loss = x.new_zeros((), dtype=torch.float32)
+        for c, b in zip(cls_outs, box_outs):
+            loss = loss + torch.sigmoid(c.float()).pow(2).mean() \
+                + b.float().pow(2).mean()
         loss.backward()
         opt.step()
         return loss
