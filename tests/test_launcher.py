@@ -111,3 +111,56 @@ def test_example_script_end_to_end(tmp_path):
     assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
     assert "TRAIN_OK" in r.stdout
     assert ck.exists()
+
+
+def test_multi_node_env_contract(tmp_path):
+    """Two msbn.run agents on one host emulate a 2-node x 2-proc job:
+    global RANK/WORLD_SIZE must come out right on every worker."""
+    script = tmp_path / "check.py"
+    script.write_text(
+        "import os, sys\n"
+        "import torch.distributed as dist\n"
+        "dist.init_process_group('gloo', init_method='env://')\n"
+        "t = __import__('torch').tensor([dist.get_rank()])\n"
+        "dist.all_reduce(t)\n"
+        "assert dist.get_world_size() == 4\n"
+        "assert t.item() == 6, t.item()\n"  # 0+1+2+3
+        "if dist.get_rank() == 0:\n"
+        "    print('MULTINODE_OK')\n"
+    )
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO
+    port = 29541
+    agents = []
+    for node in range(2):
+        agents.append(subprocess.Popen(
+            [sys.executable, "-m", "msbn.run", "--nproc_per_node=2",
+             "--nnodes=2", f"--node-rank={node}",
+             "--master-addr", "127.0.0.1", "--master-port", str(port),
+             str(script)],
+            env=env, cwd=REPO, stdout=subprocess.PIPE, text=True,
+        ))
+    outs = []
+    for a in agents:
+        out, _ = a.communicate(timeout=180)
+        outs.append(out)
+        assert a.returncode == 0, out
+    assert any("MULTINODE_OK" in o for o in outs)
+
+
+def test_standalone_flag(tmp_path):
+    script = tmp_path / "s.py"
+    script.write_text(
+        "import os\n"
+        "assert os.environ['MASTER_ADDR'] == '127.0.0.1'\n"
+        "print('STANDALONE_OK', os.environ['RANK'])\n"
+    )
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO
+    r = subprocess.run(
+        [sys.executable, "-m", "msbn.run", "--standalone",
+         "--nproc_per_node=2", str(script)],
+        capture_output=True, text=True, timeout=120, env=env, cwd=REPO,
+    )
+    assert r.returncode == 0, r.stderr
+    assert "STANDALONE_OK 0" in r.stdout and "STANDALONE_OK 1" in r.stdout
