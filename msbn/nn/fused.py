@@ -177,7 +177,22 @@ class SyncBatchNormAct2d(SyncBatchNorm):
         )
 
         if not bn_training:
-            # eval: running-stat normalize, composed (differentiable) path
+            # eval fast path: one fused kernel under no_grad (serving)
+            if input.is_cuda and not torch.is_grad_enabled():
+                from msbn import ops as _ops
+
+                input = _contig(input)
+                if residual is not None:
+                    residual = _match_layout(residual, input)
+                rm = self.running_mean.to(torch.float32)
+                invstd = torch.rsqrt(
+                    self.running_var.to(torch.float32) + self.eps
+                )
+                return _ops.batch_norm_elemt_act(
+                    input, residual, self.weight, self.bias, rm, invstd,
+                    self.relu,
+                )
+            # composed (differentiable) path
             rm = self.running_mean.to(torch.float32)
             rv = self.running_var.to(torch.float32)
             invstd = torch.rsqrt(rv + self.eps)
