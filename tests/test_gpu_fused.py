@@ -245,3 +245,21 @@ def test_graphed_step_gpu():
         )
         assert r.returncode == 0 and f"CASE {case} OK" in r.stdout, (
             r.stdout, r.stderr)
+
+
+def test_fused_eval_fast_path_gpu():
+    """Eval + no_grad routes through the fused kernel; matches the
+    differentiable composed path."""
+    torch.manual_seed(33)
+    m = msbn.nn.SyncBatchNormAct2d(16, relu=True).to(DEV)
+    m.train()
+    m(torch.randn(4, 16, 6, 6, device=DEV))
+    m.eval()
+    x = torch.randn(4, 16, 6, 6, device=DEV)
+    res = torch.randn(4, 16, 6, 6, device=DEV)
+    with torch.no_grad():
+        fast = m(x, res)
+    with torch.enable_grad():
+        xr = x.clone().requires_grad_(True)
+        composed = m(xr, res)
+    torch.testing.assert_close(fast, composed.detach(), atol=1e-5, rtol=1e-5)
