@@ -472,3 +472,42 @@ def _single_sample_body(rank):
 
 def test_single_sample_per_rank(tmp_path):
     _spawn("_single_sample_body", tmp_path)
+
+
+# --------------------------------------------------------------------------
+def _find_unused_body(rank):
+    """find_unused_parameters=True: a branch that never runs still gets
+    zero-contribution sync (DDP-level autograd-graph walk)."""
+    import msbn
+
+    class M(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.used = torch.nn.Linear(4, 4)
+            self.unused = torch.nn.Linear(4, 4)
+
+        def forward(self, x):
+            return self.used(x)
+
+    torch.manual_seed(0)
+    net = msbn.parallel.DistributedDataParallel(
+        M(), find_unused_parameters=True
+    )
+    opt = torch.optim.SGD(net.parameters(), lr=0.01)
+    for _ in range(3):
+        opt.zero_grad(set_to_none=True)
+        net(torch.full((2, 4), float(rank + 1))).sum().backward()
+        opt.step()
+    assert net.reducer.iterations() == 3
+    # used-branch grads averaged across ranks
+    g = net.module.used.weight.grad[0].clone()
+    g0 = g.clone()
+    dist.broadcast(g0, src=0)
+    assert torch.allclose(g, g0, atol=1e-6)
+    # unused branch never moved
+    assert net.module.unused.weight.grad is None or \
+        torch.all(net.module.unused.weight.grad == 0)
+
+
+def test_find_unused_parameters(tmp_path):
+    _spawn("_find_unused_body", tmp_path)

@@ -66,3 +66,28 @@ def test_ddp_logging_data(tmp_path):
     assert d["ints_map"]["num_parameter_tensors"] == 2
     assert d["ints_map"]["forward_count"] == 1
     assert d["strs_map"]["module_name"] == "Linear"
+
+
+def test_init_distributed_helper(monkeypatch, tmp_path):
+    """msbn.utils.env.init_distributed wires env:// like README step 2."""
+    import subprocess
+    import sys
+    import os as _os
+
+    repo = _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__)))
+    script = tmp_path / "e.py"
+    script.write_text(
+        "import sys; sys.path.insert(0, %r)\n"
+        "import os\n"
+        "from msbn.utils.env import init_distributed\n"
+        "os.environ['MASTER_PORT'] = '29551'\n"
+        "rank, world = init_distributed(local_rank=0, world_size=1)\n"
+        "assert (rank, world) == (0, 1)\n"
+        "import torch.distributed as dist\n"
+        "assert dist.get_backend() == 'gloo'\n"
+        "print('ENVINIT_OK')\n" % repo
+    )
+    r = subprocess.run([sys.executable, str(script)], capture_output=True,
+                       text=True, timeout=120)
+    assert r.returncode == 0, r.stderr
+    assert "ENVINIT_OK" in r.stdout
