@@ -31,6 +31,25 @@ import torch.distributed as dist
 from msbn import ops
 
 
+def _combined_view(sum_dy: torch.Tensor, sum_dy_xmu: torch.Tensor, C: int):
+    """The HIP backward-reduce op returns sum_dy / sum_dy_xmu as views of ONE
+    contiguous [2C] buffer; all_reduce that buffer directly instead of
+    cat-copying.  Falls back to cat when the two are not adjacent (CPU ref)."""
+    if (
+        sum_dy.untyped_storage().data_ptr()
+        == sum_dy_xmu.untyped_storage().data_ptr()
+        and sum_dy.data_ptr() + sum_dy.numel() * sum_dy.element_size()
+        == sum_dy_xmu.data_ptr()
+        and sum_dy.is_contiguous() and sum_dy_xmu.is_contiguous()
+    ):
+        base = sum_dy.new_empty(0)
+        base.set_(
+            sum_dy.untyped_storage(), sum_dy.storage_offset(), (2 * C,), (1,)
+        )
+        return base, False
+    return torch.cat([sum_dy, sum_dy_xmu]), True
+
+
 def _is_nccl_like(process_group) -> bool:
     try:
         return dist.get_backend(process_group) in ("nccl", "hccl")
@@ -199,14 +218,15 @@ class SyncBatchNormFunction(torch.autograd.Function):
             )
             if need_input_g:
                 if world_size > 1:
-                    combined = torch.cat([sum_dy, sum_dy_xmu])
+                    combined, copied = _combined_view(sum_dy, sum_dy_xmu, C)
                     dist.all_reduce(
                         combined, dist.ReduceOp.SUM, group=process_group
                     )
                     from msbn.utils.logging import comm_log
                     comm_log.record("all_reduce", combined.numel() * 4,
                                     f"syncbn bwd C={C}")
-                    sum_dy, sum_dy_xmu = combined[:C], combined[C:]
+                    if copied:
+                        sum_dy, sum_dy_xmu = combined[:C], combined[C:]
                 grad_input = ops.batch_norm_backward_elemt(
                     grad_output, input, mean, invstd, weight,
                     sum_dy, sum_dy_xmu, count_sum,

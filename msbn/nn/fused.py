@@ -18,7 +18,8 @@ import torch.distributed as dist
 
 from msbn import ops
 from msbn.nn.batchnorm import SyncBatchNorm, _momentum_factor
-from msbn.nn.functions import _contig, _match_layout, compute_sync_stats
+from msbn.nn.functions import (_combined_view, _contig, _match_layout,
+                               compute_sync_stats)
 
 
 class SyncBatchNormActFunction(torch.autograd.Function):
@@ -122,12 +123,13 @@ class SyncBatchNormActFunction(torch.autograd.Function):
         grad_input = grad_res = None
         if need_input_g or need_res_g:
             if world_size > 1:
-                combined = torch.cat([sum_dy, sum_dy_xmu])
+                combined, copied = _combined_view(sum_dy, sum_dy_xmu, C)
                 dist.all_reduce(combined, dist.ReduceOp.SUM, group=process_group)
                 from msbn.utils.logging import comm_log
                 comm_log.record("all_reduce", combined.numel() * 4,
                                 f"syncbn bwd C={C}")
-                sum_dy, sum_dy_xmu = combined[:C], combined[C:]
+                if copied:
+                    sum_dy, sum_dy_xmu = combined[:C], combined[C:]
             if use_gm:
                 grad_input, _ = ops.batch_norm_backward_elemt_act(
                     gm, input, None, mean, invstd, weight, bias,

@@ -91,3 +91,18 @@ def test_init_distributed_helper(monkeypatch, tmp_path):
                        text=True, timeout=120)
     assert r.returncode == 0, r.stderr
     assert "ENVINIT_OK" in r.stdout
+
+
+def test_combined_view_adjacent_and_not():
+    from msbn.nn.functions import _combined_view
+
+    base = torch.arange(8, dtype=torch.float32)
+    a, b = base[:4], base[4:]
+    v, copied = _combined_view(a, b, 4)
+    assert not copied
+    v.add_(1)  # in-place through the view reaches the originals
+    assert a[0].item() == 1.0 and b[0].item() == 5.0
+    # non-adjacent -> cat fallback
+    c, d = torch.zeros(4), torch.ones(4)
+    v2, copied2 = _combined_view(c, d, 4)
+    assert copied2 and v2.shape == (8,)
