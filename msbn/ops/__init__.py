@@ -11,7 +11,7 @@ Dispatch policy:
 Op semantics: SURVEY.md §2.3 (the five stock SyncBatchNorm ATen ops).
 """
 
-from typing import Optional, Tuple
+from typing import Optional
 
 import torch
 

@@ -14,7 +14,7 @@ async RCCL all-reduce overlapped with backward — SURVEY.md §2.2
 
 import itertools
 from contextlib import contextmanager
-from typing import Optional
+
 
 import torch
 import torch.distributed as dist
