@@ -106,3 +106,18 @@ def test_combined_view_adjacent_and_not():
     c, d = torch.zeros(4), torch.ones(4)
     v2, copied2 = _combined_view(c, d, 4)
     assert copied2 and v2.shape == (8,)
+
+
+def test_dtype_sorted_grouping():
+    """Buffer/state broadcasts must group by dtype (one flat chunk per dtype,
+    not one per consecutive run: BN models interleave fp32 stats with int64
+    counters)."""
+    from msbn.parallel.distributed import DistributedDataParallel as DDP
+
+    ts = [torch.zeros(2), torch.zeros(2, dtype=torch.long), torch.zeros(2),
+          torch.zeros(2, dtype=torch.long), torch.zeros(2)]
+    out = DDP._dtype_sorted(ts)
+    kinds = [t.dtype for t in out]
+    # all fp32 first (first-seen dtype), then all int64 — 2 runs total
+    runs = 1 + sum(1 for a, b in zip(kinds, kinds[1:]) if a != b)
+    assert runs == 2, kinds
