@@ -164,3 +164,21 @@ def test_standalone_flag(tmp_path):
     )
     assert r.returncode == 0, r.stderr
     assert "STANDALONE_OK 0" in r.stdout and "STANDALONE_OK 1" in r.stdout
+
+
+def test_stock_torch_distributed_launch_interop(tmp_path):
+    """The reference's exact launcher (`python -m torch.distributed.launch`,
+    README.md:98-100) drives msbn workers unchanged — env/argv contracts
+    are byte-compatible."""
+    script = tmp_path / "worker.py"
+    script.write_text(WORKER.format(repo=REPO))
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.launch",
+         "--nproc_per_node=2", "--master_addr", "127.0.0.1",
+         "--master_port", "29561", str(script), "--ngpu=2"],
+        capture_output=True, text=True, timeout=180, env=env, cwd=REPO,
+    )
+    assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
+    assert "LAUNCH_OK" in r.stdout
