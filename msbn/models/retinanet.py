@@ -86,23 +86,21 @@ class RetinaNet(nn.Module):
         return cls_outs, box_outs
 
     def training_loss(self, images):
-        """Dense synthetic training objective: focal loss against sparse random
-        positives + smooth-L1 on the box deltas (shapes as in real training)."""
+        """Dense synthetic training objective: focal loss against sparse
+        deterministic positives + smooth-L1 on the box deltas (shapes as in
+        real training).  Targets are generated ON DEVICE (no host work in
+        the step)."""
         cls_outs, box_outs = self(images)
-        loss = images.new_zeros(())
-        g = torch.Generator(device="cpu")
-        g.manual_seed(0)
+        loss = images.new_zeros((), dtype=torch.float32)
         for c, b in zip(cls_outs, box_outs):
-            tgt = torch.zeros_like(c)
-            # ~1% synthetic positives, deterministic
-            mask = (
-                torch.rand(c.shape, generator=g, device="cpu") < 0.01
-            ).to(c.device)
-            tgt[mask] = 1.0
+            # ~1% deterministic synthetic positives, device-side
+            idx = torch.arange(c.numel(), device=c.device)
+            tgt = (idx % 97 == 0).to(torch.float32).reshape(c.shape)
             p = torch.sigmoid(c.float())
             pt = p * tgt + (1 - p) * (1 - tgt)
             focal = -((1 - pt) ** 2) * torch.log(pt.clamp_min(1e-6))
             loss = loss + focal.mean()
-            loss = loss + F.smooth_l1_loss(b.float(),
-                                           torch.zeros_like(b, dtype=torch.float))
+            loss = loss + F.smooth_l1_loss(
+                b.float(), torch.zeros_like(b, dtype=torch.float32)
+            )
         return loss
