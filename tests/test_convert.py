@@ -74,3 +74,16 @@ def test_version1_state_dict_migration():
     fresh = SyncBatchNorm(4)
     fresh._load_from_state_dict(sd, "", {"version": 1}, True, [], [], [])
     assert fresh.num_batches_tracked is not None
+
+
+def test_converts_stock_torch_syncbn():
+    """A model already converted with torch.nn.SyncBatchNorm re-converts to
+    msbn's (stock SyncBatchNorm subclasses _BatchNorm)."""
+    m = nn.Sequential(nn.Conv2d(3, 4, 1), nn.BatchNorm2d(4))
+    stock = torch.nn.SyncBatchNorm.convert_sync_batchnorm(m)
+    assert isinstance(stock[1], torch.nn.SyncBatchNorm)
+    with torch.no_grad():
+        stock[1].running_mean.fill_(0.5)
+    ours = msbn.convert_sync_batchnorm(stock)
+    assert isinstance(ours[1], SyncBatchNorm)
+    assert ours[1].running_mean[0].item() == 0.5
