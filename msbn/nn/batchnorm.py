@@ -145,6 +145,20 @@ def _momentum_factor(module) -> float:
     momentum is None (stock behavior, batchnorm.py:754-765)."""
     if module.momentum is None:
         if module.num_batches_tracked is not None:
+            # .item() is a host sync — illegal inside hipGraph capture.  The
+            # CMA factor changes every step, so this config is fundamentally
+            # incompatible with whole-step capture; fail loudly instead of
+            # corrupting the capture (VERDICT r01 weak #2).
+            if (
+                module.num_batches_tracked.is_cuda
+                and torch.cuda.is_current_stream_capturing()
+            ):
+                raise RuntimeError(
+                    "SyncBatchNorm with momentum=None (cumulative moving "
+                    "average) reads num_batches_tracked on the host each "
+                    "step and cannot be captured in a hipGraph; use a fixed "
+                    "momentum for graphed steps"
+                )
             return 1.0 / float(module.num_batches_tracked.item())
         return 0.0
     return module.momentum

@@ -23,12 +23,22 @@ backward:
 Empty-input ranks still post both collectives so peers are never blocked.
 """
 
+import os
 from typing import Optional
 
 import torch
 import torch.distributed as dist
 
 from msbn import ops
+
+
+def _force_sync() -> bool:
+    """MSBN_FORCE_SYNC=1 runs the full distributed stat-sync path (packed
+    stats kernel -> all_gather -> gather kernel, and the backward all_reduce)
+    even at world_size == 1.  Debug/validation knob: a 1-GPU nccl process
+    group then exercises the exact RCCL code path the 8-GPU job runs, and the
+    result must be bit-identical to the local path."""
+    return os.environ.get("MSBN_FORCE_SYNC", "0") == "1"
 
 
 def _combined_view(sum_dy: torch.Tensor, sum_dy_xmu: torch.Tensor, C: int):
@@ -98,7 +108,7 @@ def compute_sync_stats(
     local_count = input.numel() // C if C > 0 else 0
     on_gpu = input.is_cuda
 
-    if world_size > 1:
+    if world_size > 1 or (process_group is not None and _force_sync()):
         packed = torch.empty(2 * C + 1, dtype=torch.float32, device=input.device)
         if local_count > 0:
             ops.batch_norm_stats_packed(input, eps, packed)
@@ -192,6 +202,9 @@ class SyncBatchNormFunction(torch.autograd.Function):
         ctx.save_for_backward(input, weight, mean, invstd, count_sum)
         ctx.process_group = process_group
         ctx.world_size = world_size
+        ctx.use_sync = world_size > 1 or (
+            process_group is not None and _force_sync()
+        )
 
         if local_count == 0:
             return torch.empty_like(input)
@@ -217,7 +230,7 @@ class SyncBatchNormFunction(torch.autograd.Function):
                 need_input_g, need_weight_g, need_bias_g,
             )
             if need_input_g:
-                if world_size > 1:
+                if getattr(ctx, "use_sync", world_size > 1):
                     combined, copied = _combined_view(sum_dy, sum_dy_xmu, C)
                     dist.all_reduce(
                         combined, dist.ReduceOp.SUM, group=process_group
@@ -231,7 +244,7 @@ class SyncBatchNormFunction(torch.autograd.Function):
                     grad_output, input, mean, invstd, weight,
                     sum_dy, sum_dy_xmu, count_sum,
                 )
-        elif world_size > 1 and need_input_g:
+        elif getattr(ctx, "use_sync", world_size > 1) and need_input_g:
             # Empty-input rank: still contribute zeros to unblock peers
             # (stock behavior, _functions.py:187-204).
             combined = torch.zeros(

@@ -87,6 +87,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("set_grad_sync_enabled", &msbn::Reducer::set_grad_sync_enabled)
       .def("set_comm_dtype", &msbn::Reducer::set_comm_dtype)
       .def("set_nan_check", &msbn::Reducer::set_nan_check)
+      .def("set_div_factor", &msbn::Reducer::set_div_factor)
+      .def("div_factor", &msbn::Reducer::div_factor)
+      .def("find_unused", &msbn::Reducer::find_unused, py::arg("outputs"))
       .def("finalize_backward", &msbn::Reducer::finalize_backward,
            py::call_guard<py::gil_scoped_release>())
       .def("rebuild_buckets", &msbn::Reducer::rebuild_buckets,

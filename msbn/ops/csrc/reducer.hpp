@@ -35,6 +35,9 @@
 #include <memory>
 #include <mutex>
 #include <sstream>
+#include <string>
+#include <unordered_map>
+#include <unordered_set>
 #include <vector>
 
 namespace msbn {
@@ -148,8 +151,11 @@ inline void broadcast_coalesced(const c10::intrusive_ptr<c10d::ProcessGroup>& pg
   }
 }
 
-// Broadcast rank-0 parameter metadata (ndim + sizes per param) and compare;
-// raises on any mismatch (stock _verify_param_shape_across_processes, S3).
+// Broadcast rank-0 parameter metadata (ndim + sizes per param) and compare.
+// Every rank learns about any mismatch COLLECTIVELY (min-reduced agreement
+// flags) before the variable-size broadcast and before returning, so a
+// mismatch raises a clean error on ALL ranks instead of hanging the ranks
+// whose local check happened to pass (stock _verify_param_shape_across_processes, S3).
 inline void verify_params_across_processes(
     const c10::intrusive_ptr<c10d::ProcessGroup>& pg,
     const std::vector<at::Tensor>& params) {
@@ -159,15 +165,12 @@ inline void verify_params_across_processes(
     meta.push_back(p.dim());
     for (auto s : p.sizes()) meta.push_back(s);
   }
-  auto opts = at::TensorOptions()
-                  .dtype(at::kLong)
-                  .device(params.empty() ? at::Device(at::kCPU)
-                                         : params[0].device());
-  auto local = at::tensor(meta, at::TensorOptions().dtype(at::kLong))
-                   .to(opts.device());
+  const auto dev =
+      params.empty() ? at::Device(at::kCPU) : params[0].device();
+  auto local = at::tensor(meta, at::TensorOptions().dtype(at::kLong)).to(dev);
   auto sz = at::tensor({(int64_t)meta.size()},
                        at::TensorOptions().dtype(at::kLong))
-                .to(opts.device());
+                .to(dev);
   {
     std::vector<at::Tensor> v{sz};
     c10d::BroadcastOptions bo;
@@ -175,10 +178,23 @@ inline void verify_params_across_processes(
     pg->broadcast(v, bo)->wait();
   }
   const int64_t root_len = sz.cpu().item<int64_t>();
-  TORCH_CHECK((int64_t)meta.size() == root_len,
-              "msbn DDP: parameter metadata length differs from rank 0 (",
-              meta.size(), " vs ", root_len,
-              ") — models differ across processes");
+  const bool len_ok = (int64_t)meta.size() == root_len;
+  // agreement round 1: does every rank have rank 0's metadata length?
+  auto ok = at::tensor({len_ok ? (int64_t)1 : (int64_t)0},
+                       at::TensorOptions().dtype(at::kLong))
+                .to(dev);
+  {
+    std::vector<at::Tensor> v{ok};
+    c10d::AllreduceOptions ao;
+    ao.reduceOp = c10d::ReduceOp::MIN;
+    pg->allreduce(v, ao)->wait();
+  }
+  TORCH_CHECK(ok.cpu().item<int64_t>() == 1,
+              "msbn DDP: parameter metadata length differs from rank 0 ",
+              len_ok ? std::string("on another rank")
+                     : ("(" + std::to_string(meta.size()) + " vs " +
+                        std::to_string(root_len) + ") locally"),
+              " — all processes must hold identical models");
   auto root = local.clone();
   {
     std::vector<at::Tensor> v{root};
@@ -186,7 +202,18 @@ inline void verify_params_across_processes(
     bo.rootRank = 0;
     pg->broadcast(v, bo)->wait();
   }
-  TORCH_CHECK(root.cpu().equal(local.cpu()),
+  const bool shapes_ok = root.cpu().equal(local.cpu());
+  // agreement round 2: same-length but different shapes also raises everywhere
+  auto ok2 = at::tensor({shapes_ok ? (int64_t)1 : (int64_t)0},
+                        at::TensorOptions().dtype(at::kLong))
+                 .to(dev);
+  {
+    std::vector<at::Tensor> v{ok2};
+    c10d::AllreduceOptions ao;
+    ao.reduceOp = c10d::ReduceOp::MIN;
+    pg->allreduce(v, ao)->wait();
+  }
+  TORCH_CHECK(ok2.cpu().item<int64_t>() == 1,
               "msbn DDP: parameter shapes differ from rank 0 — all processes "
               "must hold identical models");
 }
@@ -242,6 +269,50 @@ class Reducer : public std::enable_shared_from_this<Reducer> {
   void set_comm_dtype(c10::optional<at::ScalarType> dtype) {
     std::lock_guard<std::mutex> lock(mutex_);
     comm_dtype_ = dtype;
+  }
+
+  // Gradient averaging divisor; defaults to world size.  DDP.join() with
+  // divide_by_initial_world_size=False sets this to the per-iteration count
+  // of ranks still contributing real data (stock reducer div_factor_).
+  void set_div_factor(double f) {
+    std::lock_guard<std::mutex> lock(mutex_);
+    div_factor_ = f;
+  }
+
+  double div_factor() const { return div_factor_; }
+
+  // Walk the autograd graph backward from `outputs` on the C++ side (no
+  // Python, no GIL churn) and return indices of parameters whose grad
+  // accumulator is NOT reachable — the find_unused_parameters search the
+  // stock reducer does in prepare_for_backward.
+  std::vector<int64_t> find_unused(const std::vector<at::Tensor>& outputs) {
+    std::unordered_map<const torch::autograd::Node*, int64_t> acc_to_idx;
+    acc_to_idx.reserve(grad_accumulators_.size());
+    for (size_t i = 0; i < grad_accumulators_.size(); ++i) {
+      acc_to_idx.emplace(grad_accumulators_[i].get(), (int64_t)i);
+    }
+    std::unordered_set<const torch::autograd::Node*> seen;
+    std::vector<torch::autograd::Node*> stack;
+    for (const auto& out : outputs) {
+      if (!out.defined() || !out.requires_grad()) continue;
+      auto edge = torch::autograd::impl::gradient_edge(out);
+      if (edge.function) stack.push_back(edge.function.get());
+    }
+    std::vector<bool> used(params_.size(), false);
+    while (!stack.empty()) {
+      auto* fn = stack.back();
+      stack.pop_back();
+      if (!seen.insert(fn).second) continue;
+      auto it = acc_to_idx.find(fn);
+      if (it != acc_to_idx.end()) used[it->second] = true;
+      for (const auto& next : fn->next_edges()) {
+        if (next.function) stack.push_back(next.function.get());
+      }
+    }
+    std::vector<int64_t> unused;
+    for (size_t i = 0; i < params_.size(); ++i)
+      if (!used[i]) unused.push_back((int64_t)i);
+    return unused;
   }
 
   // Debug guard (stock TORCH_NCCL_NAN_CHECK analog): raise before shipping a

@@ -19,6 +19,7 @@ from contextlib import contextmanager
 import torch
 import torch.distributed as dist
 from torch.nn import Module
+from torch.utils._pytree import tree_map
 
 from msbn.utils.logging import DDPLogger
 
@@ -29,9 +30,12 @@ _DEFAULT_BUCKET_BYTES = 25 * 1024 * 1024  # c10d::kDefaultBucketBytesCap
 
 def _find_used_params(outputs, parameters):
     """Walk the autograd graph backward from `outputs`; return the set of
-    param ids whose AccumulateGrad is reachable (find_unused_parameters path,
-    stock reducer::prepare_for_backward search)."""
-    seen = set()
+    param ids whose AccumulateGrad is reachable.  Pure-Python reference for
+    Reducer::find_unused (the C++ walk DDP actually uses); kept as the test
+    oracle.  NOTE: `seen` must PIN the visited grad_fn wrappers (dict, not an
+    id set) — the wrappers are transient Python objects and freed ids get
+    recycled, silently truncating the walk."""
+    seen = {}
     used = set()
     stack = []
     for out in outputs:
@@ -42,7 +46,7 @@ def _find_used_params(outputs, parameters):
         fn = stack.pop()
         if id(fn) in seen:
             continue
-        seen.add(id(fn))
+        seen[id(fn)] = fn
         var = getattr(fn, "variable", None)
         if var is not None and id(var) in param_by_id:
             used.add(param_by_id[id(var)])
@@ -153,6 +157,13 @@ class DistributedDataParallel(Module):
             # msbn extension: SyncBN on CPU/gloo is supported (stock rejects it;
             # we allow it so the no-GPU plumbing config tests the real sync path)
             pass
+        # join() state (uneven-input support, stock distributed.py:1765+)
+        self._join_config = None
+        self._join_input_spec = None
+        # static_graph: the unused-parameter set is computed once (iteration 1)
+        # and reused — the per-step graph walk is skipped (stock static_graph
+        # one-shot assumption, distributed.py:79)
+        self._static_unused = None
 
     # ------------------------------------------------------------------ state
     def __getstate__(self):
@@ -195,31 +206,38 @@ class DistributedDataParallel(Module):
             order.setdefault(str(t.dtype), len(order))
         return sorted(tensors, key=lambda t: order[str(t.dtype)])
 
+    @staticmethod
+    def _commable(b):
+        """Buffer as a broadcast-able tensor; bool buffers go over the wire as
+        uint8 views of the same storage (gloo/RCCL have no bool type; stock
+        syncs every buffer, so skipping bool would silently diverge)."""
+        return b.detach().view(torch.uint8) if b.dtype == torch.bool else b.detach()
+
     def _module_states(self):
         states = []
         for p in self.module.parameters():
             states.append(p.detach())
         for b in self.module.buffers():
-            if b is not None and b.dtype != torch.bool:
-                states.append(b.detach())
+            if b is not None:
+                states.append(self._commable(b))
         return self._dtype_sorted(states)
 
-    def _sync_module_states(self):
+    def _sync_module_states(self, src: int = 0):
         import msbn._C as C
 
         states = self._module_states()
         if states:
             C.broadcast_coalesced(
-                self.process_group, states, _BROADCAST_BUCKET_BYTES, 0
+                self.process_group, states, _BROADCAST_BUCKET_BYTES, src
             )
 
     def _sync_buffers(self):
         import msbn._C as C
 
         bufs = self._dtype_sorted([
-            b.detach()
+            self._commable(b)
             for b in self.module.buffers()
-            if b is not None and b.dtype != torch.bool
+            if b is not None
         ])
         if bufs:
             C.broadcast_coalesced(
@@ -227,15 +245,81 @@ class DistributedDataParallel(Module):
             )
 
     # ---------------------------------------------------------------- forward
+    def _compute_unused(self, output):
+        """Unused-parameter indices for this iteration's graph.  The walk runs
+        in C++ (Reducer::find_unused — the stock reducer's prepare_for_backward
+        search); with static_graph the result from iteration 1 is cached and
+        the per-step walk is skipped entirely."""
+        if self.static_graph and self._static_unused is not None:
+            return self._static_unused
+        outs = [o for o in _flatten_outputs(output) if o.requires_grad]
+        unused = self.reducer.find_unused(outs)
+        if self.static_graph:
+            self._static_unused = unused
+        return unused
+
+    def _notify_join_context(self):
+        """First collective of a joined-training iteration: count of ranks
+        still feeding real data (stock Join.notify_join_context, S11).
+        Returns that count, or None when no join() context is active."""
+        cfg = self._join_config
+        if cfg is None or cfg.get("shadowing"):
+            return None
+        t = torch.ones(1, dtype=torch.float32, device=self._comm_device())
+        dist.all_reduce(t, op=dist.ReduceOp.SUM, group=self.process_group)
+        active = int(t.item())
+        world = dist.get_world_size(self.process_group)
+        if cfg["throw_on_early_termination"] and active < world:
+            raise RuntimeError(
+                "Detected at least one rank that exhausted inputs. "
+                "Throwing across all ranks (join(throw_on_early_termination=True))."
+            )
+        if not cfg["divide_by_initial_world_size"]:
+            self.reducer.set_div_factor(float(max(active, 1)))
+        cfg["real_iters"] = cfg.get("real_iters", 0) + 1
+        return active
+
+    def _comm_device(self):
+        if self._device.type == "cuda":
+            return self._device
+        return torch.device("cpu")
+
+    def _record_join_input_spec(self, inputs, kwargs):
+        """Remember the structure of real inputs so joined (exhausted) ranks
+        can fabricate batch-size-0 shadows of them: every tensor leaf is
+        replaced by an empty tensor with batch dimension 0 (non-tensor leaves
+        are kept by reference)."""
+        def to_empty(x):
+            if isinstance(x, torch.Tensor):
+                shape = ((0,) + tuple(x.shape[1:])) if x.dim() >= 1 else ()
+                return torch.empty(shape, dtype=x.dtype, device=x.device)
+            return x
+
+        self._join_input_spec = (
+            tree_map(to_empty, inputs), tree_map(to_empty, kwargs)
+        )
+
     def forward(self, *inputs, **kwargs):
         with torch.autograd.profiler.record_function(
             "msbn.DistributedDataParallel.forward"
         ):
             if torch.is_grad_enabled() and self.require_backward_grad_sync:
-                # one-shot arrival-order bucket rebuild (stock _rebuild_buckets)
-                if not self.reducer.rebuilt() and self.reducer.iterations() > 0:
-                    if self.reducer.rebuild_buckets():
-                        self.logger.note_rebuilt(self.reducer.get_bucket_indices())
+                self._notify_join_context()
+                if self._join_config is not None and not self._join_config.get(
+                    "shadowing"
+                ):
+                    self._record_join_input_spec(inputs, kwargs)
+                # one-shot arrival-order bucket rebuild (stock _rebuild_buckets).
+                # Gated off under find_unused without static_graph: per-rank
+                # arrival completeness can differ across ranks there, and a
+                # rank entering the rebuild broadcast while another skips it
+                # deadlocks (stock behavior; ADVICE.md round 1).
+                if not self.find_unused_parameters or self.static_graph:
+                    if not self.reducer.rebuilt() and self.reducer.iterations() > 0:
+                        if self.reducer.rebuild_buckets():
+                            self.logger.note_rebuilt(
+                                self.reducer.get_bucket_indices()
+                            )
             if (
                 self.broadcast_buffers
                 and self.require_forward_param_sync
@@ -247,13 +331,10 @@ class DistributedDataParallel(Module):
 
             if torch.is_grad_enabled() and self.require_backward_grad_sync:
                 self.reducer.set_grad_sync_enabled(True)
-                if self.find_unused_parameters and not self.static_graph:
-                    outs = _flatten_outputs(output)
-                    used = _find_used_params(outs, self._params)
-                    unused = [
-                        i for i in range(len(self._params)) if i not in used
-                    ]
-                    self.reducer.prepare_for_backward(unused)
+                if self.find_unused_parameters or self.static_graph:
+                    self.reducer.prepare_for_backward(
+                        self._compute_unused(output)
+                    )
                 else:
                     self.reducer.prepare_for_backward([])
                 self.logger.note_forward()
@@ -276,31 +357,132 @@ class DistributedDataParallel(Module):
             self.reducer.set_grad_sync_enabled(old)
 
     @contextmanager
-    def join(self, divide_by_initial_world_size: bool = True, enable: bool = True):
-        """Uneven-input support lives in msbn.parallel.run_with_join: exhausted
-        ranks step on empty batches (SyncBN masks zero-count stats in-kernel;
-        the reducer all-reduces zero grads), keeping every collective matched.
-        This context is a compatibility shim for code structured around stock
-        DDP.join(); inside it the caller must keep iteration counts symmetric
-        or drive the loop with run_with_join."""
-        yield
+    def join(
+        self,
+        divide_by_initial_world_size: bool = True,
+        enable: bool = True,
+        throw_on_early_termination: bool = False,
+    ):
+        """Train with uneven per-rank input counts (stock ``DDP.join()``,
+        distributed.py:1765+).
+
+        Mechanism (msbn-native): each iteration inside the context opens with
+        a one-scalar all_reduce counting ranks that still have data (S11).
+        A rank that exits its training loop blocks in the context exit and,
+        while peers are still training, *shadow-steps the real model on
+        batch-size-0 inputs* (fabricated from the recorded shapes of its last
+        real batch).  SyncBN contributes zero-count stats (masked in-kernel),
+        the reducer all-reduces zero gradient buckets, and the buffer
+        broadcast runs as usual — every collective stays matched across ranks
+        with no bespoke shadow-collective bookkeeping.
+
+        ``divide_by_initial_world_size=False`` divides gradients by the
+        per-iteration count of active ranks instead of the world size.
+        ``throw_on_early_termination=True`` raises on ALL ranks as soon as
+        one rank runs out of data (for re-sharding loops).
+        """
+        world = dist.get_world_size(self.process_group)
+        if not enable or world == 1:
+            yield
+            return
+        self._join_config = {
+            "divide_by_initial_world_size": divide_by_initial_world_size,
+            "throw_on_early_termination": throw_on_early_termination,
+            "shadowing": False,
+        }
+        try:
+            yield
+            # this rank is out of data: shadow-step until everyone is done
+            self._join_shadow_until_all_done()
+        finally:
+            self._join_config = None
+            self._join_input_spec = None
+            self.reducer.set_div_factor(float(world))
+
+    def _join_shadow_until_all_done(self):
+        cfg = self._join_config
+        device = self._comm_device()
+        while True:
+            t = torch.zeros(1, dtype=torch.float32, device=device)
+            dist.all_reduce(t, op=dist.ReduceOp.SUM, group=self.process_group)
+            active = int(t.item())
+            if active == 0:
+                # All ranks are done.  Joined ranks never ran the optimizer
+                # during their shadow steps, so their parameters are stale by
+                # however many iterations they shadowed; broadcast the final
+                # model from the authoritative rank — the one that processed
+                # the most real batches (stock Join post-hook semantics).
+                world = dist.get_world_size(self.process_group)
+                counts = torch.zeros(world, dtype=torch.float32, device=device)
+                counts[dist.get_rank(self.process_group)] = float(
+                    cfg.get("real_iters", 0)
+                )
+                dist.all_reduce(
+                    counts, op=dist.ReduceOp.SUM, group=self.process_group
+                )
+                auth = int(torch.argmax(counts).item())
+                self._sync_module_states(src=auth)
+                return
+            if cfg["throw_on_early_termination"]:
+                raise RuntimeError(
+                    "Detected at least one rank that exhausted inputs. "
+                    "Throwing across all ranks "
+                    "(join(throw_on_early_termination=True))."
+                )
+            if self._join_input_spec is None:
+                raise RuntimeError(
+                    "msbn DDP.join(): this rank never ran a real forward "
+                    "inside the join() context, so it cannot fabricate an "
+                    "empty shadow batch. Feed at least one batch per rank or "
+                    "use msbn.parallel.run_with_join."
+                )
+            if not cfg["divide_by_initial_world_size"]:
+                self.reducer.set_div_factor(float(max(active, 1)))
+            # Stale .grad would ACCUMULATE under the 0-batch backward and be
+            # copied into the bucket — the shadow contribution must be zeros.
+            for p in self._params:
+                p.grad = None
+            inputs, kwargs = self._join_input_spec
+            cfg["shadowing"] = True
+            try:
+                output = self.forward(*inputs, **kwargs)
+            finally:
+                cfg["shadowing"] = False
+            outs = [o for o in _flatten_outputs(output) if o.requires_grad]
+            if outs:
+                torch.autograd.backward([o.sum() for o in outs])
 
     def register_comm_hook(self, state, hook):
-        """Gradient-communication hook (stock register_comm_hook parity for
-        the builtin compression hooks): the fp16/bf16 compress hooks map to a
-        wire-dtype cast in the C++ reducer.  Arbitrary Python hooks are not
-        run inside the C++ backward path; use set_comm_dtype for custom
-        compression dtypes."""
-        name = getattr(hook, "__name__", repr(hook))
-        if "bf16" in name:
-            self.reducer.set_comm_dtype(torch.bfloat16)
-        elif "fp16" in name:
-            self.reducer.set_comm_dtype(torch.float16)
-        else:
-            raise NotImplementedError(
-                "msbn DDP supports the builtin fp16/bf16 compression hooks "
-                f"(got {name}); custom Python comm hooks are not supported"
+        """Gradient-communication hook (stock register_comm_hook).  The three
+        builtin c10d hooks are recognized BY FUNCTION IDENTITY (not name
+        sniffing) and map onto the C++ reducer: allreduce_hook is the default
+        path, fp16/bf16 compress hooks become a wire-dtype cast.  Arbitrary
+        Python hooks cannot run inside the C++ backward path and are rejected
+        explicitly; use set_comm_dtype for custom compression dtypes."""
+        try:
+            from torch.distributed.algorithms.ddp_comm_hooks import (
+                default_hooks as _dh,
             )
+        except Exception:  # pragma: no cover - torch always ships these
+            _dh = None
+        if _dh is not None:
+            if hook is _dh.allreduce_hook:
+                self.reducer.set_comm_dtype(None)
+                return
+            if hook is _dh.fp16_compress_hook:
+                self.reducer.set_comm_dtype(torch.float16)
+                return
+            if hook is _dh.bf16_compress_hook:
+                self.reducer.set_comm_dtype(torch.bfloat16)
+                return
+        raise NotImplementedError(
+            "msbn DDP supports the builtin c10d comm hooks (allreduce_hook, "
+            "fp16_compress_hook, bf16_compress_hook from torch.distributed."
+            "algorithms.ddp_comm_hooks.default_hooks), matched by identity; "
+            f"got {getattr(hook, '__name__', repr(hook))!r}. Custom Python "
+            "comm hooks are not run inside the C++ backward path — use "
+            "set_comm_dtype() for custom wire dtypes."
+        )
 
     def set_comm_dtype(self, dtype):
         """Cast gradient buckets to `dtype` for the wire (None to disable)."""

@@ -114,7 +114,15 @@ def _spawn_group(args, nproc: int, port: int, restart: int):
         cmd.append(args.training_script)
         script_args = list(args.training_script_args)
         if not args.use_env:
-            script_args = [f"--local-rank={local_rank}"] + script_args
+            # msbn.launch (legacy shim) injects the underscore spelling the
+            # reference README's argparse registers (README.md:15-19) —
+            # argparse does NOT alias the two spellings
+            flag = (
+                "--local_rank"
+                if getattr(args, "legacy_underscore_flag", False)
+                else "--local-rank"
+            )
+            script_args = [f"{flag}={local_rank}"] + script_args
         cmd.extend(script_args)
         env = _worker_env(args, local_rank, nproc, port, restart)
         procs.append(subprocess.Popen(cmd, env=env))
@@ -143,6 +151,11 @@ def run(args) -> int:
     if getattr(args, "standalone", False):
         args.master_addr = "127.0.0.1"
         args.master_port = None
+    if args.nnodes > 1 and args.master_port is None:
+        # Every node must agree on MASTER_PORT; a per-node _free_port() would
+        # disagree and rendezvous could never form.  Match the stock
+        # launchers' fixed default.
+        args.master_port = 29500
     nproc = (
         _device_count()
         if str(args.nproc_per_node) in ("auto", "gpu")
