@@ -12,8 +12,10 @@ One JSON line is printed by rank 0 (contract in the project brief).
 """
 
 import argparse
+import contextlib
 import json
 import os
+import sys
 import time
 
 import torch
@@ -102,81 +104,102 @@ def main():
                else torch.contiguous_format)
 
     torch.manual_seed(1234)
-    if args.stock:
-        model = msbn.models.resnet50() if args.model == "resnet50" else \
-            msbn.models.resnet18()
-        if distributed:
-            model = torch.nn.SyncBatchNorm.convert_sync_batchnorm(model)
-        model = model.to(device)
-        if dtype == torch.bfloat16:
-            model = cast_bf16_keep_bn_fp32(model)
+    # Whole-step hipGraph capture (single OR multi GPU; msbn impl only —
+    # stock DDP needs static_graph for capture).  The model, DDP wrapper,
+    # data, warmup AND capture must share ONE side stream: the reducer's
+    # autograd hooks live on AccumulateGrad nodes created at DDP-construction
+    # time, and the engine replays them on their creation-time stream — a
+    # default-stream accumulator would launch the bucket all-reduce outside
+    # the capture (tools/nccl_world1_check.py case `graph`).
+    use_graph = (args.graph or os.environ.get("MSBN_GRAPH", "0") == "1") \
+        and use_cuda and not args.stock
+    side_stream = torch.cuda.Stream() if use_graph else None
+    stream_ctx = (
+        torch.cuda.stream(side_stream) if use_graph else contextlib.nullcontext()
+    )
+    with stream_ctx:
+        if args.stock:
+            model = msbn.models.resnet50() if args.model == "resnet50" else \
+                msbn.models.resnet18()
+            if distributed:
+                model = torch.nn.SyncBatchNorm.convert_sync_batchnorm(model)
+            model = model.to(device)
+            if dtype == torch.bfloat16:
+                model = cast_bf16_keep_bn_fp32(model)
+            if mem_fmt == torch.channels_last:
+                model = model.to(memory_format=torch.channels_last)
+            if distributed:
+                model = torch.nn.parallel.DistributedDataParallel(
+                    model, device_ids=[args.local_rank] if use_cuda else None,
+                    output_device=args.local_rank if use_cuda else None,
+                )
+        else:
+            model = build_model(args.model, mem_fmt, dtype, device, sync=True,
+                                fused=not args.no_fused)
+            if distributed:
+                model = msbn.parallel.DistributedDataParallel(
+                    model,
+                    device_ids=[args.local_rank] if use_cuda else None,
+                    output_device=args.local_rank if use_cuda else None,
+                    gradient_as_bucket_view=True,
+                    bucket_cap_mb=float(os.environ.get("MSBN_BUCKET_MB", "25")),
+                )
+        model.train()
+
+        try:  # fused foreach SGD (works on ROCm; identical for both impls)
+            opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9,
+                                  weight_decay=1e-4, fused=True)
+        except (RuntimeError, TypeError):
+            opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9,
+                                  weight_decay=1e-4)
+        loss_fn = torch.nn.CrossEntropyLoss()
+
+        bs = args.batch_size
+        x = torch.randn(bs, 3, 224, 224, device=device, dtype=dtype)
         if mem_fmt == torch.channels_last:
-            model = model.to(memory_format=torch.channels_last)
-        if distributed:
-            model = torch.nn.parallel.DistributedDataParallel(
-                model, device_ids=[args.local_rank] if use_cuda else None,
-                output_device=args.local_rank if use_cuda else None,
-            )
-    else:
-        model = build_model(args.model, mem_fmt, dtype, device, sync=True,
-                            fused=not args.no_fused)
-        if distributed:
-            model = msbn.parallel.DistributedDataParallel(
-                model,
-                device_ids=[args.local_rank] if use_cuda else None,
-                output_device=args.local_rank if use_cuda else None,
-                gradient_as_bucket_view=True,
-                bucket_cap_mb=float(os.environ.get("MSBN_BUCKET_MB", "25")),
-            )
-    model.train()
+            x = x.to(memory_format=torch.channels_last)
+        y = torch.randint(0, 1000, (bs,), device=device)
 
-    try:  # fused foreach SGD (works on ROCm; identical for both impls)
-        opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9,
-                              weight_decay=1e-4, fused=True)
-    except (RuntimeError, TypeError):
-        opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9,
-                              weight_decay=1e-4)
-    loss_fn = torch.nn.CrossEntropyLoss()
-
-    bs = args.batch_size
-    x = torch.randn(bs, 3, 224, 224, device=device, dtype=dtype)
-    if mem_fmt == torch.channels_last:
-        x = x.to(memory_format=torch.channels_last)
-    y = torch.randint(0, 1000, (bs,), device=device)
-
-    def step():
-        opt.zero_grad(set_to_none=True)
-        out = model(x)
-        loss = loss_fn(out.float(), y)
-        loss.backward()
-        opt.step()
-        return loss
-
-    graph = None
-    use_graph = args.graph or os.environ.get("MSBN_GRAPH", "0") == "1"
-    if use_graph and use_cuda and not distributed:
-        # hipGraph-captured whole train step (fwd+bwd+optimizer): kills the
-        # ~500 per-step kernel-launch round trips.  msbn's BN ops are
-        # capture-safe by design (no host syncs; zero-count masking is
-        # in-kernel).
-        for _ in range(max(args.warmup, 3)):
-            step()
-        opt.zero_grad(set_to_none=False)
-        torch.cuda.synchronize()
-        graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(graph):
+        def step():
+            opt.zero_grad(set_to_none=True)
             out = model(x)
             loss = loss_fn(out.float(), y)
             loss.backward()
             opt.step()
-            # zero grads IN-GRAPH so replays are self-contained
-            torch._foreach_zero_([p.grad for p in model.parameters()
-                                  if p.grad is not None])
-        for _ in range(args.warmup):
-            graph.replay()
-    else:
-        for _ in range(args.warmup):
+            return loss
+
+        # warmup (on the side stream when graphing: allocator steady state,
+        # MIOpen finds, RCCL communicator creation)
+        for _ in range(max(args.warmup, 3) if use_graph else args.warmup):
             step()
+        if use_graph:
+            opt.zero_grad(set_to_none=False)
+
+    graph = None
+    if use_graph:
+        # hipGraph-captured whole train step (fwd+bwd+optimizer+collectives):
+        # kills the ~500 per-step kernel-launch round trips AND the per-layer
+        # collective launch latency.  msbn's BN ops are capture-safe by
+        # design (no host syncs; zero-count masking is in-kernel).  Falls
+        # back to eager on any capture failure.
+        torch.cuda.synchronize()
+        try:
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph, stream=side_stream):
+                out = model(x)
+                loss = loss_fn(out.float(), y)
+                loss.backward()
+                opt.step()
+                # zero grads IN-GRAPH so replays are self-contained
+                torch._foreach_zero_([p.grad for p in model.parameters()
+                                      if p.grad is not None])
+            for _ in range(args.warmup):
+                graph.replay()
+        except Exception as e:  # capture unsupported for this config
+            print(f"[bench] hipGraph capture failed ({e!r}); "
+                  "falling back to eager", file=sys.stderr)
+            graph = None
+            torch.cuda.synchronize()
 
     if distributed:
         dist.barrier()

@@ -88,31 +88,49 @@ def case_graph():
 
     steps = 4
 
-    def train(model, graphed):
-        opt = torch.optim.SGD(model.parameters(), lr=0.05, momentum=0.9)
-        loss_fn = torch.nn.CrossEntropyLoss()
-        gen = torch.Generator().manual_seed(123)
-        x = torch.randn(8, 3, 8, 8, generator=gen).cuda()
-        y = torch.randint(0, 10, (8,), generator=gen).cuda()
+    def train(graphed):
+        # DDP construction, warmup AND capture all happen on ONE side
+        # stream: the Reducer's autograd hooks hang off AccumulateGrad nodes
+        # created at DDP-construction time, and the engine runs those nodes
+        # on their creation-time stream — if that is the default stream, the
+        # hook's bucket copy + RCCL all-reduce would be launched OUTSIDE the
+        # capture and segfault it (the torch input_buffer.cpp stream-mismatch
+        # warning).  Same recipe as stock DDP+CUDA-graph usage.
+        side = torch.cuda.Stream() if graphed else None
+        import contextlib
 
-        def step():
-            out = model(x)
-            loss = loss_fn(out, y)
-            loss.backward()
-            opt.step()
-            torch._foreach_zero_(
-                [p.grad for p in model.parameters() if p.grad is not None]
+        stream_ctx = (
+            torch.cuda.stream(side) if graphed else contextlib.nullcontext()
+        )
+        with stream_ctx:
+            model = make_model(seed=21)
+            model = msbn.parallel.DistributedDataParallel(
+                model, device_ids=[0], output_device=0
             )
-            return loss
+            opt = torch.optim.SGD(model.parameters(), lr=0.05, momentum=0.9)
+            loss_fn = torch.nn.CrossEntropyLoss()
+            gen = torch.Generator().manual_seed(123)
+            x = torch.randn(8, 3, 8, 8, generator=gen).cuda()
+            y = torch.randint(0, 10, (8,), generator=gen).cuda()
 
-        # warmup (allocator steady state; also primes RCCL communicators)
-        for _ in range(3):
-            opt.zero_grad(set_to_none=False)
-            step()
+            def step():
+                out = model(x)
+                loss = loss_fn(out, y)
+                loss.backward()
+                opt.step()
+                torch._foreach_zero_(
+                    [p.grad for p in model.parameters() if p.grad is not None]
+                )
+                return loss
+
+            # warmup (allocator steady state; primes RCCL communicators)
+            for _ in range(3):
+                opt.zero_grad(set_to_none=False)
+                step()
         torch.cuda.synchronize()
         if graphed:
             g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
+            with torch.cuda.graph(g, stream=side):
                 step()
             for _ in range(steps):
                 g.replay()
@@ -123,15 +141,8 @@ def case_graph():
         return torch.cat([p.detach().flatten().float()
                           for p in model.parameters()])
 
-    m1 = make_model(seed=21)
-    ddp1 = msbn.parallel.DistributedDataParallel(m1, device_ids=[0],
-                                                 output_device=0)
-    p_eager = train(ddp1, graphed=False)
-
-    m2 = make_model(seed=21)
-    ddp2 = msbn.parallel.DistributedDataParallel(m2, device_ids=[0],
-                                                 output_device=0)
-    p_graph = train(ddp2, graphed=True)
+    p_eager = train(graphed=False)
+    p_graph = train(graphed=True)
 
     diff = (p_eager - p_graph).abs().max().item()
     assert torch.isfinite(p_graph).all().item(), "non-finite params after replay"
