@@ -26,6 +26,9 @@ def main():
     p.add_argument("--warmup", type=int, default=5)
     p.add_argument("--batch-size", type=int, default=64)
     p.add_argument("--image-size", type=int, default=64)
+    p.add_argument("--stock", action="store_true",
+                   help="stock torch SyncBatchNorm+DDP comparison line "
+                        "(identical architecture and init)")
     p.add_argument("--local_rank", "--local-rank", type=int,
                    default=int(os.environ.get("LOCAL_RANK", 0)),
                    dest="local_rank")
@@ -43,15 +46,30 @@ def main():
                                 rank=rank)
 
     torch.manual_seed(7)
-    from msbn.nn import fuse_bn_act
+    if args.stock:
+        from msbn.models import convert_to_torch_batchnorm
 
-    G = fuse_bn_act(msbn.convert_sync_batchnorm(msbn.models.Generator())).to(device)
-    D = msbn.convert_sync_batchnorm(msbn.models.Discriminator()).to(device)
-    if world > 1:
-        G = msbn.parallel.DistributedDataParallel(
-            G, device_ids=[args.local_rank] if use_cuda else None)
-        D = msbn.parallel.DistributedDataParallel(
-            D, device_ids=[args.local_rank] if use_cuda else None)
+        G = convert_to_torch_batchnorm(msbn.models.Generator())
+        D = convert_to_torch_batchnorm(msbn.models.Discriminator())
+        if world > 1:
+            G = torch.nn.SyncBatchNorm.convert_sync_batchnorm(G)
+            D = torch.nn.SyncBatchNorm.convert_sync_batchnorm(D)
+        G, D = G.to(device), D.to(device)
+        if world > 1:
+            G = torch.nn.parallel.DistributedDataParallel(
+                G, device_ids=[args.local_rank] if use_cuda else None)
+            D = torch.nn.parallel.DistributedDataParallel(
+                D, device_ids=[args.local_rank] if use_cuda else None)
+    else:
+        from msbn.nn import fuse_bn_act
+
+        G = fuse_bn_act(msbn.convert_sync_batchnorm(msbn.models.Generator())).to(device)
+        D = msbn.convert_sync_batchnorm(msbn.models.Discriminator()).to(device)
+        if world > 1:
+            G = msbn.parallel.DistributedDataParallel(
+                G, device_ids=[args.local_rank] if use_cuda else None)
+            D = msbn.parallel.DistributedDataParallel(
+                D, device_ids=[args.local_rank] if use_cuda else None)
     optG = torch.optim.Adam(G.parameters(), lr=2e-4, betas=(0.5, 0.999))
     optD = torch.optim.Adam(D.parameters(), lr=2e-4, betas=(0.5, 0.999))
     bce = torch.nn.BCEWithLogitsLoss()
@@ -110,7 +128,8 @@ def main():
             "dtype": "fp32",
             "data": "synthetic",
             "config": {"model": "dcgan64", "per_gpu_batch": bs,
-                       "parallelism": f"dp{world}"},
+                       "parallelism": f"dp{world}",
+                       "impl": "stock" if args.stock else "msbn"},
         }))
     if world > 1:
         dist.destroy_process_group()

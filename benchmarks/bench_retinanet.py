@@ -29,6 +29,9 @@ def main():
     p.add_argument("--width", type=int, default=1333)
     p.add_argument("--dtype", type=str, default="bf16",
                    choices=["bf16", "fp32"])
+    p.add_argument("--stock", action="store_true",
+                   help="stock torch SyncBatchNorm+DDP comparison line "
+                        "(identical architecture and init)")
     p.add_argument("--local_rank", "--local-rank", type=int,
                    default=int(os.environ.get("LOCAL_RANK", 0)),
                    dest="local_rank")
@@ -46,7 +49,15 @@ def main():
                                 rank=rank)
 
     torch.manual_seed(11)
-    model = msbn.convert_sync_batchnorm(msbn.models.retinanet()).to(device)
+    if args.stock:
+        from msbn.models import convert_to_torch_batchnorm
+
+        model = convert_to_torch_batchnorm(msbn.models.retinanet())
+        if world > 1:
+            model = torch.nn.SyncBatchNorm.convert_sync_batchnorm(model)
+        model = model.to(device)
+    else:
+        model = msbn.convert_sync_batchnorm(msbn.models.retinanet()).to(device)
     dtype = torch.bfloat16 if (args.dtype == "bf16" and use_cuda) else \
         torch.float32
     if dtype == torch.bfloat16:
@@ -56,9 +67,13 @@ def main():
     if use_cuda:
         model = model.to(memory_format=torch.channels_last)
     if world > 1:
-        model = msbn.parallel.DistributedDataParallel(
-            model, device_ids=[args.local_rank] if use_cuda else None,
-            gradient_as_bucket_view=True)
+        if args.stock:
+            model = torch.nn.parallel.DistributedDataParallel(
+                model, device_ids=[args.local_rank] if use_cuda else None)
+        else:
+            model = msbn.parallel.DistributedDataParallel(
+                model, device_ids=[args.local_rank] if use_cuda else None,
+                gradient_as_bucket_view=True)
     opt = torch.optim.SGD(model.parameters(), lr=0.01, momentum=0.9)
 
     # pad to multiples of 32 for the FPN strides (real detectors pad too)
@@ -117,7 +132,8 @@ def main():
             "config": {"model": "retinanet_r50_fpn",
                        "image": f"3x{H}x{W}",
                        "per_gpu_batch": args.batch_size,
-                       "parallelism": f"dp{world}"},
+                       "parallelism": f"dp{world}",
+                       "impl": "stock" if args.stock else "msbn"},
         }))
     if world > 1:
         dist.destroy_process_group()
