@@ -48,14 +48,14 @@ __global__ void bn_stats_partial_nchw(const T* __restrict__ x,
     const T* row = x + (n * C + c) * S;
     if (V == 1) {
       for (int64_t s = s0 + threadIdx.x; s < s1; s += blockDim.x) {
-        float v = to_f(row[s]);
+        float v = to_f(nt_load1(&row[s]));
         a += v;
         b += (double)v * v;
       }
     } else {
       for (int64_t s = s0 + (int64_t)threadIdx.x * V; s < s1;
            s += (int64_t)blockDim.x * V) {
-        Pack<T, V> pk = *reinterpret_cast<const Pack<T, V>*>(&row[s]);
+        Pack<T, V> pk = nt_load<T, V>(&row[s]);
 #pragma unroll
         for (int k = 0; k < V; ++k) {
           float v = to_f(pk.v[k]);
@@ -90,7 +90,7 @@ __global__ void bn_stats_partial_nchw_flat(const T* __restrict__ x,
   double a = 0.0, b = 0.0;
   for (int64_t p = p0 + threadIdx.x; p < p1; p += blockDim.x) {
     const int64_t n = p / S, s = p - n * S;
-    const float v = to_f(x[(n * C + c) * S + s]);
+    const float v = to_f(nt_load1(&x[(n * C + c) * S + s]));
     a += v;
     b += (double)v * v;
   }
@@ -127,7 +127,7 @@ __global__ void bn_stats_partial_nhwc(const T* __restrict__ x,
     const int64_t r0 = (int64_t)blockIdx.y * chunk_rows;
     const int64_t r1 = i64min(r0 + chunk_rows, rows);
     for (int64_t r = r0 + rowoff; r < r1; r += rpi) {
-      Pack<T, V> pk = *reinterpret_cast<const Pack<T, V>*>(&x[r * C + c]);
+      Pack<T, V> pk = nt_load<T, V>(&x[r * C + c]);
 #pragma unroll
       for (int k = 0; k < V; ++k) {
         float v = to_f(pk.v[k]);
@@ -342,14 +342,14 @@ __global__ void bn_elemt_nchw(const T* __restrict__ x,
     const int64_t c = (e / S) % C;
     const float sc = scale[c], sh = shift[c];
     if (V == 1) {
-      float z = to_f(x[e]) * sc + sh;
-      if (RES) z += to_f(res[e]);
+      float z = to_f(nt_load1(&x[e])) * sc + sh;
+      if (RES) z += to_f(nt_load1(&res[e]));
       if (RELU) z = fmaxf(z, 0.f);
-      y[e] = from_f<T>(z);
+      nt_store1(&y[e], from_f<T>(z));
     } else {
-      Pack<T, V> px = *reinterpret_cast<const Pack<T, V>*>(&x[e]);
+      Pack<T, V> px = nt_load<T, V>(&x[e]);
       Pack<T, V> pr;
-      if (RES) pr = *reinterpret_cast<const Pack<T, V>*>(&res[e]);
+      if (RES) pr = nt_load<T, V>(&res[e]);
       Pack<T, V> py;
 #pragma unroll
       for (int k = 0; k < V; ++k) {
@@ -358,7 +358,7 @@ __global__ void bn_elemt_nchw(const T* __restrict__ x,
         if (RELU) z = fmaxf(z, 0.f);
         py.v[k] = from_f<T>(z);
       }
-      *reinterpret_cast<Pack<T, V>*>(&y[e]) = py;
+      nt_store<T, V>(&y[e], py);
     }
   }
 }
@@ -375,16 +375,16 @@ __global__ void bn_elemt_nhwc(const T* __restrict__ x,
     const int64_t e = i * V;
     const int64_t c = e % C;
     if (V == 1) {
-      float z = to_f(x[e]) * scale[c] + shift[c];
-      if (RES) z += to_f(res[e]);
+      float z = to_f(nt_load1(&x[e])) * scale[c] + shift[c];
+      if (RES) z += to_f(nt_load1(&res[e]));
       if (RELU) z = fmaxf(z, 0.f);
-      y[e] = from_f<T>(z);
+      nt_store1(&y[e], from_f<T>(z));
     } else {
-      Pack<T, V> px = *reinterpret_cast<const Pack<T, V>*>(&x[e]);
+      Pack<T, V> px = nt_load<T, V>(&x[e]);
       Pack<float, V> ps = *reinterpret_cast<const Pack<float, V>*>(&scale[c]);
       Pack<float, V> pb = *reinterpret_cast<const Pack<float, V>*>(&shift[c]);
       Pack<T, V> pr;
-      if (RES) pr = *reinterpret_cast<const Pack<T, V>*>(&res[e]);
+      if (RES) pr = nt_load<T, V>(&res[e]);
       Pack<T, V> py;
 #pragma unroll
       for (int k = 0; k < V; ++k) {
@@ -393,7 +393,7 @@ __global__ void bn_elemt_nhwc(const T* __restrict__ x,
         if (RELU) z = fmaxf(z, 0.f);
         py.v[k] = from_f<T>(z);
       }
-      *reinterpret_cast<Pack<T, V>*>(&y[e]) = py;
+      nt_store<T, V>(&y[e], py);
     }
   }
 }
@@ -428,24 +428,24 @@ __global__ void bn_bwd_reduce_partial_nchw(const T* __restrict__ dy,
     const int64_t base = (n * C + c) * S;
     if (V == 1) {
       for (int64_t s = s0 + threadIdx.x; s < s1; s += blockDim.x) {
-        float g = to_f(dy[base + s]);
-        const float xv = to_f(x[base + s]);
+        float g = to_f(nt_load1(&dy[base + s]));
+        const float xv = to_f(nt_load1(&x[base + s]));
         if (MASK) {
           float z = sc * xv + sh;
-          if (RES) z += to_f(res[base + s]);
+          if (RES) z += to_f(nt_load1(&res[base + s]));
           if (z <= 0.f) g = 0.f;
         }
-        if (GMOUT) gm_out[base + s] = from_f<T>(g);
+        if (GMOUT) nt_store1(&gm_out[base + s], from_f<T>(g));
         a += g;
         b += (double)g * (xv - m);
       }
     } else {
       for (int64_t s = s0 + (int64_t)threadIdx.x * V; s < s1;
            s += (int64_t)blockDim.x * V) {
-        Pack<T, V> pg = *reinterpret_cast<const Pack<T, V>*>(&dy[base + s]);
-        Pack<T, V> px = *reinterpret_cast<const Pack<T, V>*>(&x[base + s]);
+        Pack<T, V> pg = nt_load<T, V>(&dy[base + s]);
+        Pack<T, V> px = nt_load<T, V>(&x[base + s]);
         Pack<T, V> pr, pm;
-        if (RES) pr = *reinterpret_cast<const Pack<T, V>*>(&res[base + s]);
+        if (RES) pr = nt_load<T, V>(&res[base + s]);
 #pragma unroll
         for (int k = 0; k < V; ++k) {
           float g = to_f(pg.v[k]);
@@ -459,7 +459,7 @@ __global__ void bn_bwd_reduce_partial_nchw(const T* __restrict__ dy,
           a += g;
           b += (double)g * (xv - m);
         }
-        if (GMOUT) *reinterpret_cast<Pack<T, V>*>(&gm_out[base + s]) = pm;
+        if (GMOUT) nt_store<T, V>(&gm_out[base + s], pm);
       }
     }
   }
@@ -492,14 +492,14 @@ __global__ void bn_bwd_reduce_partial_nchw_flat(
   for (int64_t p = p0 + threadIdx.x; p < p1; p += blockDim.x) {
     const int64_t n = p / S, s = p - n * S;
     const int64_t e = (n * C + c) * S + s;
-    float g = to_f(dy[e]);
-    const float xv = to_f(x[e]);
+    float g = to_f(nt_load1(&dy[e]));
+    const float xv = to_f(nt_load1(&x[e]));
     if (MASK) {
       float z = sc * xv + sh;
-      if (RES) z += to_f(res[e]);
+      if (RES) z += to_f(nt_load1(&res[e]));
       if (z <= 0.f) g = 0.f;
     }
-    if (GMOUT) gm_out[e] = from_f<T>(g);
+    if (GMOUT) nt_store1(&gm_out[e], from_f<T>(g));
     a += g;
     b += (double)g * (xv - m);
   }
@@ -549,10 +549,10 @@ __global__ void bn_bwd_reduce_partial_nhwc(const T* __restrict__ dy,
     const int64_t r0 = (int64_t)blockIdx.y * chunk_rows;
     const int64_t r1 = i64min(r0 + chunk_rows, rows);
     for (int64_t r = r0 + rowoff; r < r1; r += rpi) {
-      Pack<T, V> pg = *reinterpret_cast<const Pack<T, V>*>(&dy[r * C + c]);
-      Pack<T, V> px = *reinterpret_cast<const Pack<T, V>*>(&x[r * C + c]);
+      Pack<T, V> pg = nt_load<T, V>(&dy[r * C + c]);
+      Pack<T, V> px = nt_load<T, V>(&x[r * C + c]);
       Pack<T, V> pr, pm;
-      if (RES) pr = *reinterpret_cast<const Pack<T, V>*>(&res[r * C + c]);
+      if (RES) pr = nt_load<T, V>(&res[r * C + c]);
 #pragma unroll
       for (int k = 0; k < V; ++k) {
         float g = to_f(pg.v[k]);
@@ -566,7 +566,7 @@ __global__ void bn_bwd_reduce_partial_nhwc(const T* __restrict__ dy,
         a[k] += g;
         b[k] += (double)g * (xv - m[k]);
       }
-      if (GMOUT) *reinterpret_cast<Pack<T, V>*>(&gm_out[r * C + c]) = pm;
+      if (GMOUT) nt_store<T, V>(&gm_out[r * C + c], pm);
     }
   }
   // LDS layout [k][tid] with separate sum/sumsq planes: lane l of a wave
@@ -655,20 +655,20 @@ __global__ void bn_bwd_elemt_nchw(const T* __restrict__ dy,
     const float sc = MASK ? scale[c] : 0.f;
     const float sh = MASK ? shift[c] : 0.f;
     if (V == 1) {
-      float g = to_f(dy[e]);
-      const float xv = to_f(x[e]);
+      float g = to_f(nt_load1(&dy[e]));
+      const float xv = to_f(nt_load1(&x[e]));
       if (MASK) {
         float z = sc * xv + sh;
-        if (RES) z += to_f(res[e]);
+        if (RES) z += to_f(nt_load1(&res[e]));
         if (z <= 0.f) g = 0.f;
       }
-      dx[e] = from_f<T>(A * g + B * xv + D);
-      if (RESG) dres[e] = from_f<T>(g);
+      nt_store1(&dx[e], from_f<T>(A * g + B * xv + D));
+      if (RESG) nt_store1(&dres[e], from_f<T>(g));
     } else {
-      Pack<T, V> pg = *reinterpret_cast<const Pack<T, V>*>(&dy[e]);
-      Pack<T, V> px = *reinterpret_cast<const Pack<T, V>*>(&x[e]);
+      Pack<T, V> pg = nt_load<T, V>(&dy[e]);
+      Pack<T, V> px = nt_load<T, V>(&x[e]);
       Pack<T, V> pr;
-      if (RES) pr = *reinterpret_cast<const Pack<T, V>*>(&res[e]);
+      if (RES) pr = nt_load<T, V>(&res[e]);
       Pack<T, V> po, pq;
 #pragma unroll
       for (int k = 0; k < V; ++k) {
@@ -682,8 +682,8 @@ __global__ void bn_bwd_elemt_nchw(const T* __restrict__ dy,
         po.v[k] = from_f<T>(A * g + B * xv + D);
         if (RESG) pq.v[k] = from_f<T>(g);
       }
-      *reinterpret_cast<Pack<T, V>*>(&dx[e]) = po;
-      if (RESG) *reinterpret_cast<Pack<T, V>*>(&dres[e]) = pq;
+      nt_store<T, V>(&dx[e], po);
+      if (RESG) nt_store<T, V>(&dres[e], pq);
     }
   }
 }
@@ -705,18 +705,18 @@ __global__ void bn_bwd_elemt_nhwc(const T* __restrict__ dy,
     const int64_t e = i * V;
     const int64_t c = e % C;
     if (V == 1) {
-      float g = to_f(dy[e]);
-      const float xv = to_f(x[e]);
+      float g = to_f(nt_load1(&dy[e]));
+      const float xv = to_f(nt_load1(&x[e]));
       if (MASK) {
         float z = scale[c] * xv + shift[c];
-        if (RES) z += to_f(res[e]);
+        if (RES) z += to_f(nt_load1(&res[e]));
         if (z <= 0.f) g = 0.f;
       }
-      dx[e] = from_f<T>(ca[c] * g + cb[c] * xv + cd[c]);
-      if (RESG) dres[e] = from_f<T>(g);
+      nt_store1(&dx[e], from_f<T>(ca[c] * g + cb[c] * xv + cd[c]));
+      if (RESG) nt_store1(&dres[e], from_f<T>(g));
     } else {
-      Pack<T, V> pg = *reinterpret_cast<const Pack<T, V>*>(&dy[e]);
-      Pack<T, V> px = *reinterpret_cast<const Pack<T, V>*>(&x[e]);
+      Pack<T, V> pg = nt_load<T, V>(&dy[e]);
+      Pack<T, V> px = nt_load<T, V>(&x[e]);
       Pack<float, V> pa = *reinterpret_cast<const Pack<float, V>*>(&ca[c]);
       Pack<float, V> pb = *reinterpret_cast<const Pack<float, V>*>(&cb[c]);
       Pack<float, V> pd = *reinterpret_cast<const Pack<float, V>*>(&cd[c]);
@@ -726,7 +726,7 @@ __global__ void bn_bwd_elemt_nhwc(const T* __restrict__ dy,
         psh = *reinterpret_cast<const Pack<float, V>*>(&shift[c]);
       }
       Pack<T, V> pr;
-      if (RES) pr = *reinterpret_cast<const Pack<T, V>*>(&res[e]);
+      if (RES) pr = nt_load<T, V>(&res[e]);
       Pack<T, V> po, pq;
 #pragma unroll
       for (int k = 0; k < V; ++k) {
@@ -740,8 +740,8 @@ __global__ void bn_bwd_elemt_nhwc(const T* __restrict__ dy,
         po.v[k] = from_f<T>(pa.v[k] * g + pb.v[k] * xv + pd.v[k]);
         if (RESG) pq.v[k] = from_f<T>(g);
       }
-      *reinterpret_cast<Pack<T, V>*>(&dx[e]) = po;
-      if (RESG) *reinterpret_cast<Pack<T, V>*>(&dres[e]) = pq;
+      nt_store<T, V>(&dx[e], po);
+      if (RESG) nt_store<T, V>(&dres[e], pq);
     }
   }
 }

@@ -45,6 +45,77 @@ __device__ __forceinline__ __half from_f<__half>(float x) {
   return __float2half(x);
 }
 
+// ------------------------------------------------ nontemporal pack load/store
+// Streaming tensors (activations / gradients: read or written exactly once
+// per launch and far larger than L2) use nontemporal accesses — the nt cache
+// bit keeps them from evicting the hot small data (per-channel coefs, fp64
+// workspace) and measured slightly ahead of cached loads at BN shapes.
+// Per-channel coefficient loads stay CACHED (reused by every block).
+template <int BYTES>
+struct NTVec;
+template <>
+struct NTVec<32> {  // 2x dwordx4 (e.g. fp32 V=8)
+  using type = int __attribute__((vector_size(32)));
+};
+template <>
+struct NTVec<16> {
+  using type = int __attribute__((vector_size(16)));
+};
+template <>
+struct NTVec<8> {
+  using type = int __attribute__((vector_size(8)));
+};
+template <>
+struct NTVec<4> {
+  using type = int;
+};
+template <>
+struct NTVec<2> {
+  using type = short;
+};
+
+template <typename T, int V>
+__device__ __forceinline__ Pack<T, V> nt_load(const T* p) {
+#ifdef MSBN_DISABLE_NT  // cached-access A/B build (tools/kernel_bench.py)
+  return *reinterpret_cast<const Pack<T, V>*>(p);
+#else
+  using VT = typename NTVec<(int)sizeof(T) * V>::type;
+  union {
+    VT v;
+    Pack<T, V> pk;
+  } u;
+  u.v = __builtin_nontemporal_load(reinterpret_cast<const VT*>(p));
+  return u.pk;
+#endif
+}
+
+template <typename T, int V>
+__device__ __forceinline__ void nt_store(T* p, const Pack<T, V>& val) {
+#ifdef MSBN_DISABLE_NT
+  *reinterpret_cast<Pack<T, V>*>(p) = val;
+#else
+  using VT = typename NTVec<(int)sizeof(T) * V>::type;
+  union {
+    VT v;
+    Pack<T, V> pk;
+  } u;
+  u.pk = val;
+  __builtin_nontemporal_store(u.v, reinterpret_cast<VT*>(p));
+#endif
+}
+
+template <typename T>
+__device__ __forceinline__ T nt_load1(const T* p) {
+  return nt_load<T, 1>(p).v[0];
+}
+
+template <typename T>
+__device__ __forceinline__ void nt_store1(T* p, T val) {
+  Pack<T, 1> pk;
+  pk.v[0] = val;
+  nt_store<T, 1>(p, pk);
+}
+
 // --------------------------------------------------------- wave64 reductions
 __device__ __forceinline__ void wave_reduce_pair(double& a, double& b) {
 #pragma unroll

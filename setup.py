@@ -29,6 +29,23 @@ ext = CUDAExtension(
     },
 )
 
+exts = [ext]
+if os.environ.get("MSBN_BUILD_NONT", "0") == "1":
+    # A/B variant with nontemporal accesses disabled (cached loads/stores),
+    # for same-box kernel comparisons: tools/kernel_bench.py --impl nont
+    exts.append(CUDAExtension(
+        name="msbn._C_nont",
+        sources=[
+            os.path.join(CSRC, "module.cpp"),
+            os.path.join(CSRC, "bn_kernels.hip"),
+        ],
+        include_dirs=[CSRC],
+        extra_compile_args={
+            "cxx": ["-O3", "-std=c++17", "-DMSBN_DISABLE_NT"],
+            "nvcc": ["-O3", "-std=c++17", "-DMSBN_DISABLE_NT"],
+        },
+    ))
+
 setup(
     name="msbn",
     version="0.1.0",
@@ -37,6 +54,6 @@ setup(
         "msbn", "msbn.nn", "msbn.ops", "msbn.parallel", "msbn.data",
         "msbn.models", "msbn.utils",
     ],
-    ext_modules=[ext],
+    ext_modules=exts,
     cmdclass={"build_ext": BuildExtension.with_options(no_python_abi_suffix=False)},
 )
