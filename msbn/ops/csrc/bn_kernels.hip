@@ -854,8 +854,26 @@ struct NhwcGrid {
 NhwcGrid nhwc_grid(int64_t rows, int64_t C, int V) {
   NhwcGrid g{};
   const int64_t cv = cdiv(C, V);
-  g.lpr = (int)std::min<int64_t>(cv, MSBN_BLOCK);
-  const int64_t ctiles = cdiv(C, (int64_t)g.lpr * V);
+  // Workspace traffic is C * nchunks * 16 B, written by the partial kernel
+  // and read back by finalize.  The old lpr = min(cv, 256) choice made
+  // ctiles = 1 for C >= 2048 (bf16), forcing nchunks ~ kTargetBlocks and a
+  // workspace up to HALF the payload (measured 1.4 TB/s effective at
+  // 512x2048x7x7 vs 6.3 TB/s ceiling).  Tile CHANNELS instead: pick lpr so
+  // ~16 channel tiles exist, which caps nchunks near kTargetBlocks/16 and
+  // keeps the workspace at a few % of the payload.  Lower bound lpr at 8
+  // lanes so each row segment stays >= 128 B (one full memory granule).
+  int64_t lpr = cdiv(cv, 16);
+  // round up to a power of two that divides the block
+  int64_t p2 = 8;
+  while (p2 < lpr) p2 <<= 1;
+  lpr = std::min<int64_t>(std::max<int64_t>(p2, 8), MSBN_BLOCK);
+  lpr = std::min(lpr, [&] {  // never exceed what C needs
+    int64_t q = 8;
+    while (q < cv) q <<= 1;
+    return std::min<int64_t>(q, MSBN_BLOCK);
+  }());
+  g.lpr = (int)lpr;
+  const int64_t ctiles = cdiv(C, lpr * V);
   const int64_t per_tile =
       std::max<int64_t>(1, kTargetBlocks / std::max<int64_t>(ctiles, 1));
   const int rpi = MSBN_BLOCK / g.lpr;

@@ -74,7 +74,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::call_guard<py::gil_scoped_release>(), py::arg("process_group"),
         py::arg("params"));
 
-  py::class_<msbn::Reducer, std::shared_ptr<msbn::Reducer>>(m, "Reducer")
+  // module_local: the A/B build (msbn._C_nont) registers the same C++ type
+  py::class_<msbn::Reducer, std::shared_ptr<msbn::Reducer>>(
+      m, "Reducer", py::module_local())
       .def(py::init<std::vector<at::Tensor>, std::vector<std::vector<int64_t>>,
                     c10::intrusive_ptr<c10d::ProcessGroup>, bool, size_t,
                     size_t>(),

@@ -35,9 +35,11 @@ if os.environ.get("MSBN_BUILD_NONT", "0") == "1":
     # for same-box kernel comparisons: tools/kernel_bench.py --impl nont
     exts.append(CUDAExtension(
         name="msbn._C_nont",
+        # dedicated translation units: sharing bn_kernels.hip would collide
+        # on the object path and silently reuse the product (NT) kernels
         sources=[
-            os.path.join(CSRC, "module.cpp"),
-            os.path.join(CSRC, "bn_kernels.hip"),
+            os.path.join(CSRC, "module_nont_ab.cpp"),
+            os.path.join(CSRC, "bn_kernels_nont_ab.hip"),
         ],
         include_dirs=[CSRC],
         extra_compile_args={
