@@ -20,7 +20,14 @@ from msbn.ops import _reference as _ref
 _C = None
 _C_IMPORT_ERROR: Optional[BaseException] = None
 try:  # built in-tree: msbn/_C*.so
-    from msbn import _C as _C  # type: ignore
+    import os as _os
+
+    if _os.environ.get("MSBN_AB_NONT", "0") == "1":
+        # A/B measurement knob: route every op through the cached-access
+        # build (no nontemporal bits).  Needs MSBN_BUILD_NONT=1 at setup.
+        from msbn import _C_nont as _C  # type: ignore
+    else:
+        from msbn import _C as _C  # type: ignore
 except Exception as e:  # pragma: no cover - exercised only when ext missing
     _C_IMPORT_ERROR = e
 
