@@ -97,6 +97,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("rebuild_buckets", &msbn::Reducer::rebuild_buckets,
            py::call_guard<py::gil_scoped_release>())
       .def("get_bucket_indices", &msbn::Reducer::get_bucket_indices)
+      .def("get_backward_stats", &msbn::Reducer::get_backward_stats)
       .def("iterations", &msbn::Reducer::iterations)
       .def("rebuilt", &msbn::Reducer::rebuilt);
 }

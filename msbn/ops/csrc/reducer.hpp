@@ -34,6 +34,7 @@
 #include <map>
 #include <memory>
 #include <mutex>
+#include <chrono>
 #include <sstream>
 #include <string>
 #include <unordered_map>
@@ -329,6 +330,11 @@ class Reducer : public std::enable_shared_from_this<Reducer> {
     if (!expect_autograd_hooks_ || !sync_enabled_) return;
     if (!param_ready_[i] && arrival_order_.size() < params_.size()) {
       arrival_order_.push_back(i);
+      const int64_t now = std::chrono::duration_cast<std::chrono::nanoseconds>(
+                              std::chrono::steady_clock::now().time_since_epoch())
+                              .count();
+      if (arrival_order_.size() == 1) first_ready_ns_ = now;
+      last_ready_ns_[i] = now;
     }
     if (!finalize_queued_) {
       finalize_queued_ = true;
@@ -432,6 +438,18 @@ class Reducer : public std::enable_shared_from_this<Reducer> {
     return out;
   }
 
+  // Per-parameter grad-ready timestamps (us since the first hook of the
+  // iteration) from the LAST completed backward — the stock reducer's
+  // backward_stats_, the data for bucket-order tuning.
+  std::vector<double> get_backward_stats() const {
+    std::vector<double> out(params_.size(), -1.0);
+    for (size_t i = 0; i < params_.size(); ++i) {
+      if (last_ready_ns_[i] >= 0)
+        out[i] = (double)(last_ready_ns_[i] - first_ready_ns_) * 1e-3;
+    }
+    return out;
+  }
+
   int64_t iterations() const { return iterations_; }
   bool rebuilt() const { return rebuilt_; }
 
@@ -467,6 +485,7 @@ class Reducer : public std::enable_shared_from_this<Reducer> {
       buckets_.push_back(std::move(b));
     }
     param_ready_.assign(params_.size(), false);
+    last_ready_ns_.assign(params_.size(), -1);
     for (size_t i = 0; i < params_.size(); ++i) {
       TORCH_CHECK(param_to_bucket_[i].first >= 0,
                   "msbn Reducer: parameter ", i, " missing from buckets");
@@ -565,6 +584,9 @@ class Reducer : public std::enable_shared_from_this<Reducer> {
 
   c10::optional<at::ScalarType> comm_dtype_;
   bool nan_check_ = false;
+
+  std::vector<int64_t> last_ready_ns_;   // steady_clock ns per param
+  int64_t first_ready_ns_ = 0;
 
   std::mutex mutex_;
   bool expect_autograd_hooks_ = false;

@@ -168,7 +168,7 @@ class DistributedDataParallel(Module):
     # ------------------------------------------------------------------ state
     def __getstate__(self):
         state = self.__dict__.copy()
-        for k in ("process_group", "reducer", "logger"):
+        for k in ("process_group", "reducer", "logger", "_h2d_stream"):
             state.pop(k, None)
         return state
 
@@ -284,6 +284,41 @@ class DistributedDataParallel(Module):
             return self._device
         return torch.device("cpu")
 
+    def _move_inputs_to_device(self, inputs, kwargs):
+        """Move CPU tensor inputs to the module's GPU on a side stream
+        overlapped with the buffer sync (stock _pre_forward's input H2D move,
+        distributed.py:1564-1571).  No-op when the module is on CPU or every
+        tensor is already on-device (the common bench/serving case)."""
+        if self._device.type != "cuda":
+            return inputs, kwargs
+        needs_move = any(
+            isinstance(t, torch.Tensor) and t.device != self._device
+            for t in itertools.chain(
+                _flatten_outputs(list(inputs)), _flatten_outputs(kwargs)
+            )
+        )
+        if not needs_move:
+            return inputs, kwargs
+        if getattr(self, "_h2d_stream", None) is None:
+            self._h2d_stream = torch.cuda.Stream(device=self._device)
+        stream = self._h2d_stream
+        current = torch.cuda.current_stream(self._device)
+        moved = []
+        with torch.cuda.stream(stream):
+            def mv(t):
+                if isinstance(t, torch.Tensor) and t.device != self._device:
+                    out = t.to(self._device, non_blocking=True)
+                    moved.append(out)
+                    return out
+                return t
+
+            inputs = tree_map(mv, inputs)
+            kwargs = tree_map(mv, kwargs)
+        current.wait_stream(stream)
+        for t in moved:
+            t.record_stream(current)
+        return inputs, kwargs
+
     def _record_join_input_spec(self, inputs, kwargs):
         """Remember the structure of real inputs so joined (exhausted) ranks
         can fabricate batch-size-0 shadows of them: every tensor leaf is
@@ -327,6 +362,7 @@ class DistributedDataParallel(Module):
             ):
                 self._sync_buffers()
 
+            inputs, kwargs = self._move_inputs_to_device(inputs, kwargs)
             output = self.module(*inputs, **kwargs)
 
             if torch.is_grad_enabled() and self.require_backward_grad_sync:
@@ -489,7 +525,14 @@ class DistributedDataParallel(Module):
         self.reducer.set_comm_dtype(dtype)
 
     def _get_ddp_logging_data(self):
-        return self.logger.data()
+        data = self.logger.data()
+        # per-param grad-ready times (us since first hook) from the last
+        # backward — the stock reducer's backward_stats_
+        stats = self.reducer.get_backward_stats()
+        if any(s >= 0 for s in stats):
+            data["ints_map"]["last_backward_span_us"] = int(max(stats))
+        data["backward_grad_ready_us"] = stats
+        return data
 
     def train(self, mode: bool = True):
         super().train(mode)

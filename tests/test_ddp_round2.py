@@ -411,6 +411,29 @@ def test_find_unused_cpp_matches_python(tmp_path):
         dist.destroy_process_group()
 
 
+# ----------------------------------------------------------- backward stats
+def _backward_stats_body(rank, world):
+    """Reducer records per-param grad-ready timestamps (stock
+    backward_stats_)."""
+    import msbn
+
+    net = msbn.parallel.DistributedDataParallel(
+        torch.nn.Sequential(torch.nn.Linear(8, 8), torch.nn.Linear(8, 4))
+    )
+    net(torch.randn(2, 8)).sum().backward()
+    data = net._get_ddp_logging_data()
+    stats = data["backward_grad_ready_us"]
+    assert len(stats) == 4 and all(s >= 0 for s in stats)
+    assert data["ints_map"]["last_backward_span_us"] >= 0
+    # later layers' grads arrive first: the last Linear's params should be
+    # ready no later than the first Linear's weight
+    assert stats[3] <= stats[0] or stats[2] <= stats[0]
+
+
+def test_backward_stats(tmp_path):
+    _spawn("_backward_stats_body", tmp_path, world=2)
+
+
 # --------------------------------------------------------------- bool buffers
 def _bool_buffer_body(rank, world):
     """bool buffers are broadcast from rank 0 like every other buffer

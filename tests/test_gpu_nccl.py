@@ -51,3 +51,9 @@ def test_nccl_world1_whole_step_hipgraph():
     """Whole train step (fwd+bwd+opt) captured in a hipGraph WITH the RCCL
     collectives inside; replays track the eager clone."""
     _run_case("graph")
+
+
+def test_nccl_world1_input_h2d_and_backward_stats():
+    """CPU inputs moved to GPU on a side stream in DDP.forward; per-param
+    grad-ready timestamps surfaced via _get_ddp_logging_data."""
+    _run_case("h2d")

@@ -150,6 +150,26 @@ def case_graph():
     print("CASE graph OK")
 
 
+def case_h2d():
+    """CPU inputs are moved to the module's GPU on a side stream inside
+    DDP.forward (stock _pre_forward input move, distributed.py:1564-1571)."""
+    import msbn
+
+    model = make_model(seed=5)
+    ddp = msbn.parallel.DistributedDataParallel(model, device_ids=[0],
+                                                output_device=0)
+    x_cpu = torch.randn(4, 3, 8, 8)
+    out = ddp(x_cpu)  # would raise device-mismatch without the move
+    assert out.is_cuda and out.shape[0] == 4
+    out.sum().backward()
+    stats = ddp._get_ddp_logging_data()
+    assert "backward_grad_ready_us" in stats
+    ready = [s for s in stats["backward_grad_ready_us"] if s >= 0]
+    assert len(ready) == len(list(p for p in model.parameters()
+                                  if p.requires_grad))
+    print("CASE h2d OK")
+
+
 def main():
     case = sys.argv[1] if len(sys.argv) > 1 else "ddp"
     torch.cuda.set_device(0)
@@ -159,6 +179,8 @@ def main():
             case_ddp()
         elif case == "graph":
             case_graph()
+        elif case == "h2d":
+            case_h2d()
         else:
             raise SystemExit(f"unknown case {case}")
     finally:
