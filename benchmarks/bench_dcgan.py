@@ -29,6 +29,9 @@ def main():
     p.add_argument("--stock", action="store_true",
                    help="stock torch SyncBatchNorm+DDP comparison line "
                         "(identical architecture and init)")
+    p.add_argument("--graph", action="store_true",
+                   help="capture the G+D step in one hipGraph (msbn only; "
+                        "the GAN regime is launch-bound)")
     p.add_argument("--local_rank", "--local-rank", type=int,
                    default=int(os.environ.get("LOCAL_RANK", 0)),
                    dest="local_rank")
@@ -46,63 +49,94 @@ def main():
                                 rank=rank)
 
     torch.manual_seed(7)
-    if args.stock:
-        from msbn.models import convert_to_torch_batchnorm
+    import contextlib
 
-        G = convert_to_torch_batchnorm(msbn.models.Generator())
-        D = convert_to_torch_batchnorm(msbn.models.Discriminator())
-        if world > 1:
-            G = torch.nn.SyncBatchNorm.convert_sync_batchnorm(G)
-            D = torch.nn.SyncBatchNorm.convert_sync_batchnorm(D)
-        G, D = G.to(device), D.to(device)
-        if world > 1:
-            G = torch.nn.parallel.DistributedDataParallel(
-                G, device_ids=[args.local_rank] if use_cuda else None)
-            D = torch.nn.parallel.DistributedDataParallel(
-                D, device_ids=[args.local_rank] if use_cuda else None)
-    else:
-        from msbn.nn import fuse_bn_act
+    use_graph = args.graph and use_cuda and not args.stock
+    side = torch.cuda.Stream() if use_graph else None
+    stream_ctx = torch.cuda.stream(side) if use_graph else contextlib.nullcontext()
+    with stream_ctx:
+        if args.stock:
+            from msbn.models import convert_to_torch_batchnorm
 
-        G = fuse_bn_act(msbn.convert_sync_batchnorm(msbn.models.Generator())).to(device)
-        D = msbn.convert_sync_batchnorm(msbn.models.Discriminator()).to(device)
-        if world > 1:
-            G = msbn.parallel.DistributedDataParallel(
-                G, device_ids=[args.local_rank] if use_cuda else None)
-            D = msbn.parallel.DistributedDataParallel(
-                D, device_ids=[args.local_rank] if use_cuda else None)
-    optG = torch.optim.Adam(G.parameters(), lr=2e-4, betas=(0.5, 0.999))
-    optD = torch.optim.Adam(D.parameters(), lr=2e-4, betas=(0.5, 0.999))
-    bce = torch.nn.BCEWithLogitsLoss()
+            G = convert_to_torch_batchnorm(msbn.models.Generator())
+            D = convert_to_torch_batchnorm(msbn.models.Discriminator())
+            if world > 1:
+                G = torch.nn.SyncBatchNorm.convert_sync_batchnorm(G)
+                D = torch.nn.SyncBatchNorm.convert_sync_batchnorm(D)
+            G, D = G.to(device), D.to(device)
+            if world > 1:
+                G = torch.nn.parallel.DistributedDataParallel(
+                    G, device_ids=[args.local_rank] if use_cuda else None)
+                D = torch.nn.parallel.DistributedDataParallel(
+                    D, device_ids=[args.local_rank] if use_cuda else None)
+        else:
+            from msbn.nn import fuse_bn_act
 
-    bs = args.batch_size
-    real = torch.randn(bs, 3, args.image_size, args.image_size, device=device)
-    ones = torch.ones(bs, 1, device=device)
-    zeros = torch.zeros(bs, 1, device=device)
+            G = fuse_bn_act(msbn.convert_sync_batchnorm(msbn.models.Generator())).to(device)
+            D = msbn.convert_sync_batchnorm(msbn.models.Discriminator()).to(device)
+            if world > 1:
+                G = msbn.parallel.DistributedDataParallel(
+                    G, device_ids=[args.local_rank] if use_cuda else None)
+                D = msbn.parallel.DistributedDataParallel(
+                    D, device_ids=[args.local_rank] if use_cuda else None)
+        optG = torch.optim.Adam(G.parameters(), lr=2e-4, betas=(0.5, 0.999))
+        optD = torch.optim.Adam(D.parameters(), lr=2e-4, betas=(0.5, 0.999))
+        bce = torch.nn.BCEWithLogitsLoss()
 
-    def step():
-        z = torch.randn(bs, 100, 1, 1, device=device)
-        # D step
-        optD.zero_grad(set_to_none=True)
-        fake = G(z)
-        d_loss = bce(D(real), ones) + bce(D(fake.detach()), zeros)
-        d_loss.backward()
-        optD.step()
-        # G step
-        optG.zero_grad(set_to_none=True)
-        g_loss = bce(D(fake), ones)
-        g_loss.backward()
-        optG.step()
-        return d_loss, g_loss
+        bs = args.batch_size
+        real = torch.randn(bs, 3, args.image_size, args.image_size, device=device)
+        ones = torch.ones(bs, 1, device=device)
+        zeros = torch.zeros(bs, 1, device=device)
 
-    for _ in range(args.warmup):
-        step()
+        def step():
+            z = torch.randn(bs, 100, 1, 1, device=device)
+            # D step
+            optD.zero_grad(set_to_none=True)
+            fake = G(z)
+            d_loss = bce(D(real), ones) + bce(D(fake.detach()), zeros)
+            d_loss.backward()
+            optD.step()
+            # G step
+            optG.zero_grad(set_to_none=True)
+            g_loss = bce(D(fake), ones)
+            g_loss.backward()
+            optG.step()
+            return d_loss, g_loss
+
+        # warmup (on the side stream when graphing)
+        for _ in range(args.warmup):
+            step()
+
+    graph = None
+    if use_graph:
+        # one hipGraph for the WHOLE G+D step (both backwards + both Adam
+        # steps + in-graph RNG for z): the GAN regime is launch-bound, this
+        # collapses ~hundreds of launches into one.  Same side-stream recipe
+        # as bench.py; eager fallback on capture failure.
+        torch.cuda.synchronize()
+        try:
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph, stream=side):
+                step()
+            for _ in range(max(args.warmup, 3)):
+                graph.replay()
+        except Exception as e:
+            print(f"[bench_dcgan] hipGraph capture failed ({e!r}); eager",
+                  file=sys.stderr)
+            graph = None
+            torch.cuda.synchronize()
+
     if world > 1:
         dist.barrier()
     if use_cuda:
         torch.cuda.synchronize()
     t0 = time.perf_counter()
-    for _ in range(args.steps):
-        step()
+    if graph is not None:
+        for _ in range(args.steps):
+            graph.replay()
+    else:
+        for _ in range(args.steps):
+            step()
     if use_cuda:
         torch.cuda.synchronize()
     if world > 1:
@@ -129,6 +163,7 @@ def main():
             "data": "synthetic",
             "config": {"model": "dcgan64", "per_gpu_batch": bs,
                        "parallelism": f"dp{world}",
+                       "hip_graph": graph is not None,
                        "impl": "stock" if args.stock else "msbn"},
         }))
     if world > 1:

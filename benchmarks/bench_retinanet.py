@@ -32,6 +32,8 @@ def main():
     p.add_argument("--stock", action="store_true",
                    help="stock torch SyncBatchNorm+DDP comparison line "
                         "(identical architecture and init)")
+    p.add_argument("--contiguous", action="store_true",
+                   help="NCHW instead of channels_last (bisect helper)")
     p.add_argument("--local_rank", "--local-rank", type=int,
                    default=int(os.environ.get("LOCAL_RANK", 0)),
                    dest="local_rank")
@@ -64,7 +66,7 @@ def main():
         from bench import cast_bf16_keep_bn_fp32
 
         model = cast_bf16_keep_bn_fp32(model)
-    if use_cuda:
+    if use_cuda and not args.contiguous:
         model = model.to(memory_format=torch.channels_last)
     if world > 1:
         if args.stock:
@@ -80,7 +82,7 @@ def main():
     H = (args.height + 31) // 32 * 32
     W = (args.width + 31) // 32 * 32
     x = torch.randn(args.batch_size, 3, H, W, device=device, dtype=dtype)
-    if use_cuda:
+    if use_cuda and not args.contiguous:
         x = x.to(memory_format=torch.channels_last)
 
     def step():
