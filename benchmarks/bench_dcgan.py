@@ -79,8 +79,14 @@ def main():
                     G, device_ids=[args.local_rank] if use_cuda else None)
                 D = msbn.parallel.DistributedDataParallel(
                     D, device_ids=[args.local_rank] if use_cuda else None)
-        optG = torch.optim.Adam(G.parameters(), lr=2e-4, betas=(0.5, 0.999))
-        optD = torch.optim.Adam(D.parameters(), lr=2e-4, betas=(0.5, 0.999))
+        # capturable Adam keeps the step/lr state on-device so the optimizer
+        # records into the hipGraph (required for --graph; tiny host cost
+        # otherwise, so only enabled when graphing)
+        adam_kw = {"capturable": True} if use_graph else {}
+        optG = torch.optim.Adam(G.parameters(), lr=2e-4, betas=(0.5, 0.999),
+                                **adam_kw)
+        optD = torch.optim.Adam(D.parameters(), lr=2e-4, betas=(0.5, 0.999),
+                                **adam_kw)
         bce = torch.nn.BCEWithLogitsLoss()
 
         bs = args.batch_size
