@@ -193,6 +193,29 @@ class SyncBatchNormFunction(torch.autograd.Function):
 
         C = int(input.shape[1])
         local_count = input.numel() // C if C > 0 else 0
+        use_sync = world_size > 1 or (
+            process_group is not None and _force_sync()
+        )
+
+        if (
+            not use_sync
+            and local_count > 0
+            and ops.bn_fused_local_eligible(
+                input, weight, bias, running_mean, running_var
+            )
+        ):
+            # single-launch small-plane path: stats + running update +
+            # normalize in ONE kernel (K10-family; GAN/small-batch regime)
+            y, mean, invstd, count_sum, _coefs = ops.batch_norm_fwd_fused_local(
+                input, None, weight, bias, eps, momentum,
+                running_mean, running_var, False,
+            )
+            ctx.save_for_backward(input, weight, mean, invstd, count_sum)
+            ctx.process_group = process_group
+            ctx.world_size = world_size
+            ctx.use_sync = False
+            ctx.local_fused = True
+            return y
 
         mean, invstd, count_sum, coefs = compute_sync_stats(
             input, eps, momentum, running_mean, running_var,
@@ -202,9 +225,8 @@ class SyncBatchNormFunction(torch.autograd.Function):
         ctx.save_for_backward(input, weight, mean, invstd, count_sum)
         ctx.process_group = process_group
         ctx.world_size = world_size
-        ctx.use_sync = world_size > 1 or (
-            process_group is not None and _force_sync()
-        )
+        ctx.use_sync = use_sync
+        ctx.local_fused = False
 
         if local_count == 0:
             return torch.empty_like(input)
@@ -223,6 +245,21 @@ class SyncBatchNormFunction(torch.autograd.Function):
         C = int(input.shape[1])
         local_count = input.numel() // C if C > 0 else 0
         grad_input = grad_weight = grad_bias = None
+
+        if getattr(ctx, "local_fused", False) and local_count > 0:
+            # single-launch backward (reduce + coefs + dx in one kernel)
+            grad_input, grad_weight, grad_bias, _ = (
+                ops.batch_norm_bwd_fused_local(
+                    grad_output, input, None, mean, invstd, weight, None,
+                    False, False, need_weight_g, need_bias_g,
+                )
+            )
+            return (
+                grad_input if need_input_g else None,
+                grad_weight if need_weight_g else None,
+                grad_bias if need_bias_g else None,
+                None, None, None, None, None, None,
+            )
 
         if local_count > 0:
             sum_dy, sum_dy_xmu, grad_weight, grad_bias = ops.batch_norm_backward_reduce(

@@ -18,8 +18,8 @@ import torch.distributed as dist
 
 from msbn import ops
 from msbn.nn.batchnorm import SyncBatchNorm, _momentum_factor
-from msbn.nn.functions import (_combined_view, _contig, _match_layout,
-                               compute_sync_stats)
+from msbn.nn.functions import (_combined_view, _contig, _force_sync,
+                               _match_layout, compute_sync_stats)
 
 
 class SyncBatchNormActFunction(torch.autograd.Function):
@@ -45,6 +45,33 @@ class SyncBatchNormActFunction(torch.autograd.Function):
             weight = weight.contiguous()
         if bias is not None:
             bias = bias.contiguous()
+
+        use_sync = world_size > 1 or (
+            process_group is not None and _force_sync()
+        )
+        if (
+            not use_sync
+            and input.numel() > 0
+            and (residual is None or residual.is_contiguous())
+            and ops.bn_fused_local_eligible(
+                input, weight, bias, running_mean, running_var
+            )
+        ):
+            # single-launch small-plane path: stats + running update +
+            # normalize(+res)(+relu) in ONE kernel (K10-family)
+            y, mean, invstd, count_sum, coefs = ops.batch_norm_fwd_fused_local(
+                input, residual, weight, bias, eps, momentum,
+                running_mean, running_var, relu,
+            )
+            ctx.save_for_backward(input, residual, weight, bias, mean,
+                                  invstd, count_sum, coefs)
+            ctx.has_coefs = True
+            ctx.local_fused = True
+            ctx.relu = relu
+            ctx.process_group = process_group
+            ctx.world_size = world_size
+            return y
+        ctx.local_fused = False
 
         # coefs ([scale|shift]) are emitted by the finalize/gather kernel in
         # the same launch and reused by forward elemt AND the two
@@ -90,6 +117,24 @@ class SyncBatchNormActFunction(torch.autograd.Function):
         need_bias_g = bias is not None and ctx.needs_input_grad[3]
 
         C = int(input.shape[1])
+
+        if getattr(ctx, "local_fused", False) and input.numel() > 0:
+            # single-launch backward (mask recompute + reduce + coefs + dx
+            # (+dres) in one kernel)
+            grad_input, grad_weight, grad_bias, grad_res = (
+                ops.batch_norm_bwd_fused_local(
+                    grad_output, input, residual, mean, invstd, weight,
+                    coefs, relu, need_res_g, need_weight_g, need_bias_g,
+                )
+            )
+            return (
+                grad_input if need_input_g else None,
+                grad_res if need_res_g else None,
+                grad_weight if need_weight_g else None,
+                grad_bias if need_bias_g else None,
+                None, None, None, None, None, None, None,
+            )
+
         # Masked-grad materialization: when the residual branch needs its
         # gradient, the reduce pass writes gm = dy*1[z>0] once; the elemt
         # pass then reads gm instead of (dy, residual) and gm itself IS the

@@ -173,6 +173,36 @@ def batch_norm_backward_elemt(
     )
 
 
+def bn_fused_local_eligible(input, weight, bias, running_mean, running_var):
+    """True when the single-launch small-plane world-1 kernels apply
+    (NCHW, plane <= 256K elems, C >= 64, fp32 stats/affine)."""
+    if not input.is_cuda or _C is None:
+        return False
+    return _C.bn_fused_local_eligible(
+        input, weight, bias, running_mean, running_var
+    )
+
+
+def batch_norm_fwd_fused_local(input, residual, weight, bias, eps, momentum,
+                               running_mean, running_var, relu):
+    """ONE kernel: local stats + running update + normalize(+res)(+relu).
+    Returns (y, mean, invstd, count[1], coefs[2C])."""
+    return _require_hip().batch_norm_fwd_fused_local(
+        input, residual, weight, bias, eps, momentum, running_mean,
+        running_var, relu,
+    )
+
+
+def batch_norm_bwd_fused_local(grad_out, input, residual, mean, invstd,
+                               weight, coefs, relu_mask, want_res_grad,
+                               weight_g, bias_g):
+    """ONE kernel: whole local BN backward (reduce + coefs + dx(+dres))."""
+    return _require_hip().batch_norm_bwd_fused_local(
+        grad_out, input, residual, mean, invstd, weight, coefs, relu_mask,
+        want_res_grad, weight_g, bias_g,
+    )
+
+
 def bn_make_coefs(mean, invstd, weight, bias):
     """Packed [scale | shift] fp32 per-channel affine (GPU); None on CPU."""
     if mean.is_cuda:

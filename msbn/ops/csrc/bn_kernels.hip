@@ -747,6 +747,128 @@ __global__ void bn_bwd_elemt_nhwc(const T* __restrict__ dy,
 }
 
 // =====================================================================
+// small-plane world-1 fused kernels (the stock K10-style single-launch
+// family, MI355X-gated): block-per-channel, TWO passes over an L2-resident
+// plane.  Used when plane = N*S is small (GAN / small-per-GPU-batch
+// regime): there the per-launch ramp of the 3-kernel
+// two-stage pipeline dominates and the plane fits cache, so the second
+// pass re-reads from L2 (plain cached accesses on purpose — no nt).
+// NCHW only; fp32 running stats / fp32 affine (host gates eligibility).
+// =====================================================================
+template <typename T, bool RELU, bool RES>
+__global__ void bn_fwd_fused_small_nchw(
+    const T* __restrict__ x, const T* __restrict__ res, T* __restrict__ y,
+    int64_t N, int64_t C, int64_t S, float eps, float momentum,
+    float* __restrict__ mean_out, float* __restrict__ invstd_out,
+    float* __restrict__ count_out, float* __restrict__ rmean,
+    float* __restrict__ rvar, const float* __restrict__ w,
+    const float* __restrict__ b, float* __restrict__ scale_out,
+    float* __restrict__ shift_out) {
+  const int64_t c = blockIdx.x;
+  const int64_t plane = N * S;
+  double a = 0.0, bb = 0.0;
+  for (int64_t p = threadIdx.x; p < plane; p += blockDim.x) {
+    const int64_t n = p / S, s = p - n * S;
+    const float v = to_f(x[(n * C + c) * S + s]);
+    a += v;
+    bb += (double)v * v;
+  }
+  __shared__ double lds[2 * (MSBN_BLOCK / MSBN_WAVE)];
+  block_reduce_pair(a, bb, lds);
+  __shared__ float sc_sh[2];
+  if (threadIdx.x == 0) {
+    const double cnt = (double)plane;
+    const double m = a / cnt;
+    double var = bb / cnt - m * m;
+    var = var > 0.0 ? var : 0.0;
+    const float istd = (float)rsqrt(var + (double)eps);
+    mean_out[c] = (float)m;
+    invstd_out[c] = istd;
+    if (c == 0 && count_out != nullptr) count_out[0] = (float)cnt;
+    if (rmean != nullptr) {
+      const double unbiased = cnt > 1.0 ? var * (cnt / (cnt - 1.0)) : var;
+      rmean[c] = (1.f - momentum) * rmean[c] + momentum * (float)m;
+      rvar[c] = (1.f - momentum) * rvar[c] + momentum * (float)unbiased;
+    }
+    const float sc = istd * (w != nullptr ? w[c] : 1.f);
+    const float sh = -(float)m * sc + (b != nullptr ? b[c] : 0.f);
+    scale_out[c] = sc;
+    shift_out[c] = sh;
+    sc_sh[0] = sc;
+    sc_sh[1] = sh;
+  }
+  __syncthreads();
+  const float sc = sc_sh[0], sh = sc_sh[1];
+  for (int64_t p = threadIdx.x; p < plane; p += blockDim.x) {
+    const int64_t n = p / S, s = p - n * S;
+    const int64_t e = (n * C + c) * S + s;
+    float z = to_f(x[e]) * sc + sh;
+    if (RES) z += to_f(res[e]);
+    if (RELU) z = fmaxf(z, 0.f);
+    y[e] = from_f<T>(z);
+  }
+}
+
+template <typename T, bool MASK, bool RES, bool RESG>
+__global__ void bn_bwd_fused_small_nchw(
+    const T* __restrict__ dy, const T* __restrict__ x,
+    const T* __restrict__ res, T* __restrict__ dx, T* __restrict__ dres,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    const float* __restrict__ scale, const float* __restrict__ shift,
+    const float* __restrict__ w, float* __restrict__ gw,
+    float* __restrict__ gb, int64_t N, int64_t C, int64_t S) {
+  const int64_t c = blockIdx.x;
+  const int64_t plane = N * S;
+  const float m = mean[c];
+  const float sc = MASK ? scale[c] : 0.f;
+  const float sh = MASK ? shift[c] : 0.f;
+  double a = 0.0, bb = 0.0;
+  for (int64_t p = threadIdx.x; p < plane; p += blockDim.x) {
+    const int64_t n = p / S, s = p - n * S;
+    const int64_t e = (n * C + c) * S + s;
+    float g = to_f(dy[e]);
+    const float xv = to_f(x[e]);
+    if (MASK) {
+      float z = sc * xv + sh;
+      if (RES) z += to_f(res[e]);
+      if (z <= 0.f) g = 0.f;
+    }
+    a += g;
+    bb += (double)g * (xv - m);
+  }
+  __shared__ double lds[2 * (MSBN_BLOCK / MSBN_WAVE)];
+  block_reduce_pair(a, bb, lds);
+  __shared__ float abd[3];
+  if (threadIdx.x == 0) {
+    const float istd = invstd[c];
+    const float n_inv = 1.f / (float)plane;
+    if (gw != nullptr) gw[c] = (float)(bb * istd);
+    if (gb != nullptr) gb[c] = (float)a;
+    const float f1 = istd * (w != nullptr ? w[c] : 1.f);
+    const float f2 = (float)a * n_inv;
+    const float f3 = istd * istd * (float)bb * n_inv;
+    abd[0] = f1;
+    abd[1] = -f1 * f3;
+    abd[2] = f1 * (f3 * m - f2);
+  }
+  __syncthreads();
+  const float A = abd[0], B = abd[1], D = abd[2];
+  for (int64_t p = threadIdx.x; p < plane; p += blockDim.x) {
+    const int64_t n = p / S, s = p - n * S;
+    const int64_t e = (n * C + c) * S + s;
+    float g = to_f(dy[e]);
+    const float xv = to_f(x[e]);
+    if (MASK) {
+      float z = sc * xv + sh;
+      if (RES) z += to_f(res[e]);
+      if (z <= 0.f) g = 0.f;
+    }
+    dx[e] = from_f<T>(A * g + B * xv + D);
+    if (RESG) dres[e] = from_f<T>(g);
+  }
+}
+
+// =====================================================================
 // host-side helpers
 // =====================================================================
 
@@ -1581,6 +1703,167 @@ at::Tensor batch_norm_backward_elemt(
   return std::get<0>(batch_norm_backward_elemt_act(
       grad_out, input, c10::nullopt, mean, invstd, weight, c10::nullopt,
       sum_dy, sum_dy_xmu, count, false, false, c10::nullopt));
+}
+
+// ---------------------------------------------------------------------------
+// small-plane world-1 fused path (single-launch forward / backward)
+// ---------------------------------------------------------------------------
+namespace {
+constexpr int64_t kFusedSmallPlaneMax = 1 << 18;  // 256K elems per channel
+constexpr int64_t kFusedSmallMinC = 64;           // >= 64 blocks on the grid
+
+bool fp32_or_absent(const c10::optional<at::Tensor>& t) {
+  return !t.has_value() || !t->defined() || t->scalar_type() == at::kFloat;
+}
+}  // namespace
+
+bool bn_fused_local_eligible(const at::Tensor& input,
+                             const c10::optional<at::Tensor>& weight,
+                             const c10::optional<at::Tensor>& bias,
+                             const c10::optional<at::Tensor>& running_mean,
+                             const c10::optional<at::Tensor>& running_var) {
+  if (!input.is_cuda() || input.dim() < 2 || input.numel() == 0) return false;
+  const auto st = input.scalar_type();
+  if (st != at::kFloat && st != at::kBFloat16 && st != at::kHalf) return false;
+  if (!input.is_contiguous()) return false;  // NCHW only
+  // a 4-D tensor that REPORTS channels-last gets the NHWC kernels instead
+  if (input.dim() >= 4 &&
+      input.suggest_memory_format() == at::MemoryFormat::ChannelsLast)
+    return false;
+  const int64_t C = input.size(1);
+  const int64_t plane = input.numel() / C;
+  if (C < kFusedSmallMinC || plane > kFusedSmallPlaneMax) return false;
+  return fp32_or_absent(weight) && fp32_or_absent(bias) &&
+         fp32_or_absent(running_mean) && fp32_or_absent(running_var);
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor>
+batch_norm_fwd_fused_local(const at::Tensor& input,
+                           const c10::optional<at::Tensor>& residual,
+                           const c10::optional<at::Tensor>& weight,
+                           const c10::optional<at::Tensor>& bias, double eps,
+                           double momentum,
+                           const c10::optional<at::Tensor>& running_mean,
+                           const c10::optional<at::Tensor>& running_var,
+                           bool relu) {
+  const int64_t C = input.size(1);
+  const int64_t N = input.size(0);
+  const int64_t S = input.numel() / std::max<int64_t>(N * C, 1);
+  auto opts = input.options().dtype(at::kFloat);
+  auto mean = at::empty({C}, opts);
+  auto invstd = at::empty({C}, opts);
+  auto count = at::empty({1}, opts);
+  auto coefs = at::empty({2 * C}, opts);
+  auto y = at::empty_like(input);
+  auto stream = cur_stream();
+  const bool has_res = residual.has_value() && residual->defined();
+  TORCH_CHECK(!has_res || residual->is_contiguous(),
+              "fused local: residual must match NCHW layout");
+  float* rm = running_mean.has_value() && running_mean->defined()
+                  ? running_mean->data_ptr<float>()
+                  : nullptr;
+  float* rv = running_var.has_value() && running_var->defined()
+                  ? running_var->data_ptr<float>()
+                  : nullptr;
+  const float* w = weight.has_value() && weight->defined()
+                       ? weight->data_ptr<float>()
+                       : nullptr;
+  const float* b =
+      bias.has_value() && bias->defined() ? bias->data_ptr<float>() : nullptr;
+  float* scale_p = coefs.data_ptr<float>();
+  float* shift_p = scale_p + C;
+  MSBN_DISPATCH_FLOAT_TYPES(input.scalar_type(), "bn_fwd_fused_local", [&] {
+    const native_t* x =
+        reinterpret_cast<const native_t*>(input.data_ptr<scalar_t>());
+    const native_t* res =
+        has_res ? reinterpret_cast<const native_t*>(
+                      residual->data_ptr<scalar_t>())
+                : nullptr;
+    native_t* yp = reinterpret_cast<native_t*>(y.data_ptr<scalar_t>());
+    auto launch = [&](auto relu_c, auto res_c) {
+      hipLaunchKernelGGL(
+          (bn_fwd_fused_small_nchw<native_t, decltype(relu_c)::value,
+                                   decltype(res_c)::value>),
+          dim3((unsigned)C), dim3(MSBN_BLOCK), 0, stream, x, res, yp, N, C, S,
+          (float)eps, (float)momentum, mean.data_ptr<float>(),
+          invstd.data_ptr<float>(), count.data_ptr<float>(), rm, rv, w, b,
+          scale_p, shift_p);
+    };
+    using T = std::true_type;
+    using F = std::false_type;
+    if (relu && has_res) launch(T{}, T{});
+    else if (relu) launch(T{}, F{});
+    else if (has_res) launch(F{}, T{});
+    else launch(F{}, F{});
+  });
+  return {y, mean, invstd, count, coefs};
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor>
+batch_norm_bwd_fused_local(const at::Tensor& grad_out, const at::Tensor& input,
+                           const c10::optional<at::Tensor>& residual,
+                           const at::Tensor& mean, const at::Tensor& invstd,
+                           const c10::optional<at::Tensor>& weight,
+                           const c10::optional<at::Tensor>& coefs,
+                           bool relu_mask, bool want_res_grad, bool weight_g,
+                           bool bias_g) {
+  const int64_t C = input.size(1);
+  const int64_t N = input.size(0);
+  const int64_t S = input.numel() / std::max<int64_t>(N * C, 1);
+  TORCH_CHECK(grad_out.is_contiguous(), "fused local bwd: grad layout");
+  TORCH_CHECK(!relu_mask || (coefs.has_value() && coefs->defined()),
+              "fused local bwd: relu mask needs coefs");
+  auto opts = input.options().dtype(at::kFloat);
+  auto dx = at::empty_like(input);
+  at::Tensor gw, gb, dres;
+  if (weight_g) gw = at::empty({C}, opts);
+  if (bias_g) gb = at::empty({C}, opts);
+  if (want_res_grad) dres = at::empty_like(grad_out);
+  auto stream = cur_stream();
+  const bool has_res = residual.has_value() && residual->defined();
+  const float* scale_p = nullptr;
+  const float* shift_p = nullptr;
+  if (relu_mask) {
+    scale_p = coefs->data_ptr<float>();
+    shift_p = scale_p + C;
+  }
+  const float* w = weight.has_value() && weight->defined()
+                       ? weight->data_ptr<float>()
+                       : nullptr;
+  MSBN_DISPATCH_FLOAT_TYPES(input.scalar_type(), "bn_bwd_fused_local", [&] {
+    const native_t* dy =
+        reinterpret_cast<const native_t*>(grad_out.data_ptr<scalar_t>());
+    const native_t* x =
+        reinterpret_cast<const native_t*>(input.data_ptr<scalar_t>());
+    const native_t* res =
+        (relu_mask && has_res)
+            ? reinterpret_cast<const native_t*>(residual->data_ptr<scalar_t>())
+            : nullptr;
+    native_t* dxp = reinterpret_cast<native_t*>(dx.data_ptr<scalar_t>());
+    native_t* drp = want_res_grad
+                        ? reinterpret_cast<native_t*>(dres.data_ptr<scalar_t>())
+                        : nullptr;
+    auto launch = [&](auto mask_c, auto res_c, auto resg_c) {
+      hipLaunchKernelGGL(
+          (bn_bwd_fused_small_nchw<native_t, decltype(mask_c)::value,
+                                   decltype(res_c)::value,
+                                   decltype(resg_c)::value>),
+          dim3((unsigned)C), dim3(MSBN_BLOCK), 0, stream, dy, x, res, dxp, drp,
+          mean.data_ptr<float>(), invstd.data_ptr<float>(), scale_p, shift_p,
+          w, weight_g ? gw.data_ptr<float>() : nullptr,
+          bias_g ? gb.data_ptr<float>() : nullptr, N, C, S);
+    };
+    using T = std::true_type;
+    using F = std::false_type;
+    const bool mres = relu_mask && has_res;
+    if (relu_mask && mres && want_res_grad) launch(T{}, T{}, T{});
+    else if (relu_mask && mres) launch(T{}, T{}, F{});
+    else if (relu_mask && want_res_grad) launch(T{}, F{}, T{});
+    else if (relu_mask) launch(T{}, F{}, F{});
+    else if (want_res_grad) launch(F{}, F{}, T{});
+    else launch(F{}, F{}, F{});
+  });
+  return {dx, gw, gb, dres};
 }
 
 }  // namespace msbn

@@ -101,6 +101,37 @@ batch_norm_backward_reduce_act(
     bool weight_g, bool bias_g, const c10::optional<at::Tensor>& coefs_in,
     const c10::optional<at::Tensor>& gm_out);
 
+// ---- small-plane world-1 fused path (single launch fwd / bwd) -------------
+// Stock K10-style block-per-channel kernels gated to small NCHW planes
+// (GAN / small-per-GPU-batch regime): one kernel does stats + running
+// update + normalize(+res)(+relu); one kernel does the whole backward.
+bool bn_fused_local_eligible(const at::Tensor& input,
+                             const c10::optional<at::Tensor>& weight,
+                             const c10::optional<at::Tensor>& bias,
+                             const c10::optional<at::Tensor>& running_mean,
+                             const c10::optional<at::Tensor>& running_var);
+
+// returns (y, mean, invstd, count[1], coefs[2C])
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor>
+batch_norm_fwd_fused_local(const at::Tensor& input,
+                           const c10::optional<at::Tensor>& residual,
+                           const c10::optional<at::Tensor>& weight,
+                           const c10::optional<at::Tensor>& bias, double eps,
+                           double momentum,
+                           const c10::optional<at::Tensor>& running_mean,
+                           const c10::optional<at::Tensor>& running_var,
+                           bool relu);
+
+// returns (dx, grad_weight?, grad_bias?, dres?)
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor>
+batch_norm_bwd_fused_local(const at::Tensor& grad_out, const at::Tensor& input,
+                           const c10::optional<at::Tensor>& residual,
+                           const at::Tensor& mean, const at::Tensor& invstd,
+                           const c10::optional<at::Tensor>& weight,
+                           const c10::optional<at::Tensor>& coefs,
+                           bool relu_mask, bool want_res_grad, bool weight_g,
+                           bool bias_g);
+
 // returns (grad_input, grad_residual-or-undefined)
 std::tuple<at::Tensor, at::Tensor> batch_norm_backward_elemt_act(
     const at::Tensor& grad_out, const at::Tensor& input,

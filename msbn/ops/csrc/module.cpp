@@ -62,6 +62,19 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("relu_mask"), py::arg("want_res_grad"),
         py::arg("coefs") = py::none());
 
+  m.def("bn_fused_local_eligible", &msbn::bn_fused_local_eligible,
+        py::arg("input"), py::arg("weight"), py::arg("bias"),
+        py::arg("running_mean"), py::arg("running_var"));
+  m.def("batch_norm_fwd_fused_local", &msbn::batch_norm_fwd_fused_local,
+        py::arg("input"), py::arg("residual"), py::arg("weight"),
+        py::arg("bias"), py::arg("eps"), py::arg("momentum"),
+        py::arg("running_mean"), py::arg("running_var"), py::arg("relu"));
+  m.def("batch_norm_bwd_fused_local", &msbn::batch_norm_bwd_fused_local,
+        py::arg("grad_out"), py::arg("input"), py::arg("residual"),
+        py::arg("mean"), py::arg("invstd"), py::arg("weight"),
+        py::arg("coefs"), py::arg("relu_mask"), py::arg("want_res_grad"),
+        py::arg("weight_g"), py::arg("bias_g"));
+
   // ---- DDP machinery ----
   m.def("compute_bucket_assignment_by_size",
         &msbn::compute_bucket_assignment_by_size, py::arg("tensors"),
