@@ -1709,8 +1709,12 @@ at::Tensor batch_norm_backward_elemt(
 // small-plane world-1 fused path (single-launch forward / backward)
 // ---------------------------------------------------------------------------
 namespace {
-constexpr int64_t kFusedSmallPlaneMax = 1 << 18;  // 256K elems per channel
-constexpr int64_t kFusedSmallMinC = 64;           // >= 64 blocks on the grid
+// Measured crossover (tools/fused_local_bench.py, gpurun_out/flb_*.log):
+// block-per-channel wins up to plane ~8K elems (128bs x 8x8 and smaller);
+// beyond that a single block serializes too much plane work and the
+// two-stage chip-filling pipeline wins (3.4x at plane 32K).
+constexpr int64_t kFusedSmallPlaneMax = 8192;
+constexpr int64_t kFusedSmallMinC = 64;  // >= 64 blocks on the grid
 
 bool fp32_or_absent(const c10::optional<at::Tensor>& t) {
   return !t.has_value() || !t->defined() || t->scalar_type() == at::kFloat;
