@@ -755,23 +755,33 @@ __global__ void bn_bwd_elemt_nhwc(const T* __restrict__ dy,
 // pass re-reads from L2 (plain cached accesses on purpose — no nt).
 // NCHW only; fp32 running stats / fp32 affine (host gates eligibility).
 // =====================================================================
-template <typename T, bool RELU, bool RES>
+// V-wide flat indexing: pack pp covers elements [pp*V, pp*V+V) of the
+// channel plane; S % V == 0 (host-gated) keeps every pack inside one row.
+// int32 arithmetic throughout (plane <= 32K elems by gate).
+template <typename T, int V, bool RELU, bool RES>
 __global__ void bn_fwd_fused_small_nchw(
     const T* __restrict__ x, const T* __restrict__ res, T* __restrict__ y,
-    int64_t N, int64_t C, int64_t S, float eps, float momentum,
+    int N, int64_t C, int S, float eps, float momentum,
     float* __restrict__ mean_out, float* __restrict__ invstd_out,
     float* __restrict__ count_out, float* __restrict__ rmean,
     float* __restrict__ rvar, const float* __restrict__ w,
     const float* __restrict__ b, float* __restrict__ scale_out,
     float* __restrict__ shift_out) {
   const int64_t c = blockIdx.x;
-  const int64_t plane = N * S;
+  const int plane = N * S;
+  const int packs = plane / V;
   double a = 0.0, bb = 0.0;
-  for (int64_t p = threadIdx.x; p < plane; p += blockDim.x) {
-    const int64_t n = p / S, s = p - n * S;
-    const float v = to_f(x[(n * C + c) * S + s]);
-    a += v;
-    bb += (double)v * v;
+  for (int pp = threadIdx.x; pp < packs; pp += blockDim.x) {
+    const int e = pp * V;
+    const int n = e / S, s = e - n * S;
+    const T* row = x + ((int64_t)n * C + c) * S + s;
+    Pack<T, V> pk = *reinterpret_cast<const Pack<T, V>*>(row);
+#pragma unroll
+    for (int k = 0; k < V; ++k) {
+      const float v = to_f(pk.v[k]);
+      a += v;
+      bb += (double)v * v;
+    }
   }
   __shared__ double lds[2 * (MSBN_BLOCK / MSBN_WAVE)];
   block_reduce_pair(a, bb, lds);
@@ -799,42 +809,59 @@ __global__ void bn_fwd_fused_small_nchw(
   }
   __syncthreads();
   const float sc = sc_sh[0], sh = sc_sh[1];
-  for (int64_t p = threadIdx.x; p < plane; p += blockDim.x) {
-    const int64_t n = p / S, s = p - n * S;
-    const int64_t e = (n * C + c) * S + s;
-    float z = to_f(x[e]) * sc + sh;
-    if (RES) z += to_f(res[e]);
-    if (RELU) z = fmaxf(z, 0.f);
-    y[e] = from_f<T>(z);
+  for (int pp = threadIdx.x; pp < packs; pp += blockDim.x) {
+    const int e = pp * V;
+    const int n = e / S, s = e - n * S;
+    const int64_t base = ((int64_t)n * C + c) * S + s;
+    Pack<T, V> px = *reinterpret_cast<const Pack<T, V>*>(&x[base]);
+    Pack<T, V> pr, py;
+    if (RES) pr = *reinterpret_cast<const Pack<T, V>*>(&res[base]);
+#pragma unroll
+    for (int k = 0; k < V; ++k) {
+      float z = to_f(px.v[k]) * sc + sh;
+      if (RES) z += to_f(pr.v[k]);
+      if (RELU) z = fmaxf(z, 0.f);
+      py.v[k] = from_f<T>(z);
+    }
+    *reinterpret_cast<Pack<T, V>*>(&y[base]) = py;
   }
 }
 
-template <typename T, bool MASK, bool RES, bool RESG>
+template <typename T, int V, bool MASK, bool RES, bool RESG>
 __global__ void bn_bwd_fused_small_nchw(
     const T* __restrict__ dy, const T* __restrict__ x,
     const T* __restrict__ res, T* __restrict__ dx, T* __restrict__ dres,
     const float* __restrict__ mean, const float* __restrict__ invstd,
     const float* __restrict__ scale, const float* __restrict__ shift,
     const float* __restrict__ w, float* __restrict__ gw,
-    float* __restrict__ gb, int64_t N, int64_t C, int64_t S) {
+    float* __restrict__ gb, int N, int64_t C, int S) {
   const int64_t c = blockIdx.x;
-  const int64_t plane = N * S;
+  const int plane = N * S;
+  const int packs = plane / V;
   const float m = mean[c];
   const float sc = MASK ? scale[c] : 0.f;
   const float sh = MASK ? shift[c] : 0.f;
   double a = 0.0, bb = 0.0;
-  for (int64_t p = threadIdx.x; p < plane; p += blockDim.x) {
-    const int64_t n = p / S, s = p - n * S;
-    const int64_t e = (n * C + c) * S + s;
-    float g = to_f(dy[e]);
-    const float xv = to_f(x[e]);
-    if (MASK) {
-      float z = sc * xv + sh;
-      if (RES) z += to_f(res[e]);
-      if (z <= 0.f) g = 0.f;
+  for (int pp = threadIdx.x; pp < packs; pp += blockDim.x) {
+    const int e = pp * V;
+    const int n = e / S, s = e - n * S;
+    const int64_t base = ((int64_t)n * C + c) * S + s;
+    Pack<T, V> pg = *reinterpret_cast<const Pack<T, V>*>(&dy[base]);
+    Pack<T, V> px = *reinterpret_cast<const Pack<T, V>*>(&x[base]);
+    Pack<T, V> pr;
+    if (MASK && RES) pr = *reinterpret_cast<const Pack<T, V>*>(&res[base]);
+#pragma unroll
+    for (int k = 0; k < V; ++k) {
+      float g = to_f(pg.v[k]);
+      const float xv = to_f(px.v[k]);
+      if (MASK) {
+        float z = sc * xv + sh;
+        if (RES) z += to_f(pr.v[k]);
+        if (z <= 0.f) g = 0.f;
+      }
+      a += g;
+      bb += (double)g * (xv - m);
     }
-    a += g;
-    bb += (double)g * (xv - m);
   }
   __shared__ double lds[2 * (MSBN_BLOCK / MSBN_WAVE)];
   block_reduce_pair(a, bb, lds);
@@ -853,18 +880,28 @@ __global__ void bn_bwd_fused_small_nchw(
   }
   __syncthreads();
   const float A = abd[0], B = abd[1], D = abd[2];
-  for (int64_t p = threadIdx.x; p < plane; p += blockDim.x) {
-    const int64_t n = p / S, s = p - n * S;
-    const int64_t e = (n * C + c) * S + s;
-    float g = to_f(dy[e]);
-    const float xv = to_f(x[e]);
-    if (MASK) {
-      float z = sc * xv + sh;
-      if (RES) z += to_f(res[e]);
-      if (z <= 0.f) g = 0.f;
+  for (int pp = threadIdx.x; pp < packs; pp += blockDim.x) {
+    const int e = pp * V;
+    const int n = e / S, s = e - n * S;
+    const int64_t base = ((int64_t)n * C + c) * S + s;
+    Pack<T, V> pg = *reinterpret_cast<const Pack<T, V>*>(&dy[base]);
+    Pack<T, V> px = *reinterpret_cast<const Pack<T, V>*>(&x[base]);
+    Pack<T, V> pr, po, pq;
+    if (MASK && RES) pr = *reinterpret_cast<const Pack<T, V>*>(&res[base]);
+#pragma unroll
+    for (int k = 0; k < V; ++k) {
+      float g = to_f(pg.v[k]);
+      const float xv = to_f(px.v[k]);
+      if (MASK) {
+        float z = sc * xv + sh;
+        if (RES) z += to_f(pr.v[k]);
+        if (z <= 0.f) g = 0.f;
+      }
+      po.v[k] = from_f<T>(A * g + B * xv + D);
+      if (RESG) pq.v[k] = from_f<T>(g);
     }
-    dx[e] = from_f<T>(A * g + B * xv + D);
-    if (RESG) dres[e] = from_f<T>(g);
+    *reinterpret_cast<Pack<T, V>*>(&dx[base]) = po;
+    if (RESG) *reinterpret_cast<Pack<T, V>*>(&dres[base]) = pq;
   }
 }
 
@@ -1735,8 +1772,14 @@ bool bn_fused_local_eligible(const at::Tensor& input,
       input.suggest_memory_format() == at::MemoryFormat::ChannelsLast)
     return false;
   const int64_t C = input.size(1);
+  const int64_t N = input.size(0);
   const int64_t plane = input.numel() / C;
-  if (C < kFusedSmallMinC || plane > kFusedSmallPlaneMax) return false;
+  const int64_t S = plane / std::max<int64_t>(N, 1);
+  const int vmax = 16 / (int)input.element_size();
+  const bool vec = (S % vmax) == 0;
+  // vectorized blocks chew 4-8x more plane per cycle -> higher crossover
+  const int64_t lim = vec ? 4 * kFusedSmallPlaneMax : kFusedSmallPlaneMax;
+  if (C < kFusedSmallMinC || plane > lim) return false;
   return fp32_or_absent(weight) && fp32_or_absent(bias) &&
          fp32_or_absent(running_mean) && fp32_or_absent(running_var);
 }
@@ -1784,21 +1827,25 @@ batch_norm_fwd_fused_local(const at::Tensor& input,
                       residual->data_ptr<scalar_t>())
                 : nullptr;
     native_t* yp = reinterpret_cast<native_t*>(y.data_ptr<scalar_t>());
-    auto launch = [&](auto relu_c, auto res_c) {
-      hipLaunchKernelGGL(
-          (bn_fwd_fused_small_nchw<native_t, decltype(relu_c)::value,
-                                   decltype(res_c)::value>),
-          dim3((unsigned)C), dim3(MSBN_BLOCK), 0, stream, x, res, yp, N, C, S,
-          (float)eps, (float)momentum, mean.data_ptr<float>(),
-          invstd.data_ptr<float>(), count.data_ptr<float>(), rm, rv, w, b,
-          scale_p, shift_p);
-    };
-    using T = std::true_type;
-    using F = std::false_type;
-    if (relu && has_res) launch(T{}, T{});
-    else if (relu) launch(T{}, F{});
-    else if (has_res) launch(F{}, T{});
-    else launch(F{}, F{});
+    constexpr int VMAX = 16 / (int)sizeof(native_t);
+    const int v = pick_v<native_t>(x, res, yp, S);
+    MSBN_DISPATCH_V(v, VMAX, [&] {
+      auto launch = [&](auto relu_c, auto res_c) {
+        hipLaunchKernelGGL(
+            (bn_fwd_fused_small_nchw<native_t, VV, decltype(relu_c)::value,
+                                     decltype(res_c)::value>),
+            dim3((unsigned)C), dim3(MSBN_BLOCK), 0, stream, x, res, yp,
+            (int)N, C, (int)S, (float)eps, (float)momentum,
+            mean.data_ptr<float>(), invstd.data_ptr<float>(),
+            count.data_ptr<float>(), rm, rv, w, b, scale_p, shift_p);
+      };
+      using T = std::true_type;
+      using F = std::false_type;
+      if (relu && has_res) launch(T{}, T{});
+      else if (relu) launch(T{}, F{});
+      else if (has_res) launch(F{}, T{});
+      else launch(F{}, F{});
+    });
   });
   return {y, mean, invstd, count, coefs};
 }
@@ -1847,25 +1894,31 @@ batch_norm_bwd_fused_local(const at::Tensor& grad_out, const at::Tensor& input,
     native_t* drp = want_res_grad
                         ? reinterpret_cast<native_t*>(dres.data_ptr<scalar_t>())
                         : nullptr;
-    auto launch = [&](auto mask_c, auto res_c, auto resg_c) {
-      hipLaunchKernelGGL(
-          (bn_bwd_fused_small_nchw<native_t, decltype(mask_c)::value,
-                                   decltype(res_c)::value,
-                                   decltype(resg_c)::value>),
-          dim3((unsigned)C), dim3(MSBN_BLOCK), 0, stream, dy, x, res, dxp, drp,
-          mean.data_ptr<float>(), invstd.data_ptr<float>(), scale_p, shift_p,
-          w, weight_g ? gw.data_ptr<float>() : nullptr,
-          bias_g ? gb.data_ptr<float>() : nullptr, N, C, S);
-    };
-    using T = std::true_type;
-    using F = std::false_type;
-    const bool mres = relu_mask && has_res;
-    if (relu_mask && mres && want_res_grad) launch(T{}, T{}, T{});
-    else if (relu_mask && mres) launch(T{}, T{}, F{});
-    else if (relu_mask && want_res_grad) launch(T{}, F{}, T{});
-    else if (relu_mask) launch(T{}, F{}, F{});
-    else if (want_res_grad) launch(F{}, F{}, T{});
-    else launch(F{}, F{}, F{});
+    constexpr int VMAX = 16 / (int)sizeof(native_t);
+    const int v0 = pick_v<native_t>(dy, x, res, S);
+    const int v1 = pick_v<native_t>(dxp, drp, nullptr, S);
+    const int v = std::min(v0, v1);
+    MSBN_DISPATCH_V(v, VMAX, [&] {
+      auto launch = [&](auto mask_c, auto res_c, auto resg_c) {
+        hipLaunchKernelGGL(
+            (bn_bwd_fused_small_nchw<native_t, VV, decltype(mask_c)::value,
+                                     decltype(res_c)::value,
+                                     decltype(resg_c)::value>),
+            dim3((unsigned)C), dim3(MSBN_BLOCK), 0, stream, dy, x, res, dxp,
+            drp, mean.data_ptr<float>(), invstd.data_ptr<float>(), scale_p,
+            shift_p, w, weight_g ? gw.data_ptr<float>() : nullptr,
+            bias_g ? gb.data_ptr<float>() : nullptr, (int)N, C, (int)S);
+      };
+      using T = std::true_type;
+      using F = std::false_type;
+      const bool mres = relu_mask && has_res;
+      if (relu_mask && mres && want_res_grad) launch(T{}, T{}, T{});
+      else if (relu_mask && mres) launch(T{}, T{}, F{});
+      else if (relu_mask && want_res_grad) launch(T{}, F{}, T{});
+      else if (relu_mask) launch(T{}, F{}, F{});
+      else if (want_res_grad) launch(F{}, F{}, T{});
+      else launch(F{}, F{}, F{});
+    });
   });
   return {dx, gw, gb, dres};
 }
