@@ -1778,7 +1778,8 @@ bool bn_fused_local_eligible(const at::Tensor& input,
   const int vmax = 16 / (int)input.element_size();
   const bool vec = (S % vmax) == 0;
   // vectorized blocks chew 4-8x more plane per cycle -> higher crossover
-  const int64_t lim = vec ? 4 * kFusedSmallPlaneMax : kFusedSmallPlaneMax;
+  // (measured: win at 16K, slight loss at 32K — gpurun_out/flb2.log)
+  const int64_t lim = vec ? 2 * kFusedSmallPlaneMax : kFusedSmallPlaneMax;
   if (C < kFusedSmallMinC || plane > lim) return false;
   return fp32_or_absent(weight) && fp32_or_absent(bias) &&
          fp32_or_absent(running_mean) && fp32_or_absent(running_var);

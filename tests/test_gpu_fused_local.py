@@ -108,8 +108,12 @@ def test_eligibility_gates():
     x_cl = torch.randn(8, 64, 8, 8, device=DEV).to(
         memory_format=torch.channels_last)
     assert not ops.bn_fused_local_eligible(x_cl, None, None, None, None)
-    x_bigplane = torch.randn(64, 64, 16, 16, device=DEV)  # plane 16K
+    x_vec16k = torch.randn(64, 64, 16, 16, device=DEV)  # plane 16K, S%8==0
+    assert ops.bn_fused_local_eligible(x_vec16k, None, None, None, None)
+    x_bigplane = torch.randn(64, 64, 32, 32, device=DEV)  # plane 64K
     assert not ops.bn_fused_local_eligible(x_bigplane, None, None, None, None)
+    x_odd12k = torch.randn(81, 64, 11, 13, device=DEV)  # plane 11.6K, S odd
+    assert not ops.bn_fused_local_eligible(x_odd12k, None, None, None, None)
     x_smallc = torch.randn(8, 16, 8, 8, device=DEV)
     assert not ops.bn_fused_local_eligible(x_smallc, None, None, None, None)
     x_ok = torch.randn(8, 64, 8, 8, device=DEV)
