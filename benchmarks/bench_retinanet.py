@@ -32,8 +32,11 @@ def main():
     p.add_argument("--stock", action="store_true",
                    help="stock torch SyncBatchNorm+DDP comparison line "
                         "(identical architecture and init)")
-    p.add_argument("--contiguous", action="store_true",
-                   help="NCHW instead of channels_last (bisect helper)")
+    p.add_argument("--channels-last", action="store_true",
+                   help="channels_last memory format (NCHW is the measured "
+                        "best for detection shapes on this stack, and stock "
+                        "torch-ROCm SEGFAULTS on bf16+channels_last here — "
+                        "profiles/r02_kernel_polish.md §6)")
     p.add_argument("--local_rank", "--local-rank", type=int,
                    default=int(os.environ.get("LOCAL_RANK", 0)),
                    dest="local_rank")
@@ -66,7 +69,7 @@ def main():
         from bench import cast_bf16_keep_bn_fp32
 
         model = cast_bf16_keep_bn_fp32(model)
-    if use_cuda and not args.contiguous:
+    if use_cuda and args.channels_last:
         model = model.to(memory_format=torch.channels_last)
     if world > 1:
         if args.stock:
@@ -82,7 +85,7 @@ def main():
     H = (args.height + 31) // 32 * 32
     W = (args.width + 31) // 32 * 32
     x = torch.randn(args.batch_size, 3, H, W, device=device, dtype=dtype)
-    if use_cuda and not args.contiguous:
+    if use_cuda and args.channels_last:
         x = x.to(memory_format=torch.channels_last)
 
     def step():
@@ -135,6 +138,8 @@ def main():
                        "image": f"3x{H}x{W}",
                        "per_gpu_batch": args.batch_size,
                        "parallelism": f"dp{world}",
+                       "memory_format": "channels_last" if args.channels_last
+                       else "contiguous",
                        "impl": "stock" if args.stock else "msbn"},
         }))
     if world > 1:

@@ -60,3 +60,30 @@ def test_bench_distributed_launch_cpu():
     assert out["n_gpus"] == 2
     assert out["config"]["parallelism"] == "dp2"
     assert out["config"]["global_batch"] == 4
+
+
+def test_bench_multirank_cpu_gloo(tmp_path):
+    """bench.py end-to-end under torch.distributed.run with 2 CPU/gloo ranks
+    (VERDICT r01 next-round #1 done-criterion): the distributed branch —
+    msbn DDP wrap, barriers, MAX-over-ranks timing, single JSON line."""
+    import json
+    import subprocess
+    import sys
+
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29617", os.path.join(REPO, "bench.py"),
+         "--gpus", "2", "--steps", "1", "--warmup", "0",
+         "--batch-size", "2", "--model", "resnet18", "--dtype", "fp32",
+         "--memory-format", "contiguous"],
+        capture_output=True, text=True, timeout=600, cwd=REPO, env=env,
+    )
+    assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    d = json.loads(line)
+    assert d["n_gpus"] == 2 and d["config"]["parallelism"] == "dp2"
+    assert d["config"]["global_batch"] == 4
+    assert d["value"] > 0
