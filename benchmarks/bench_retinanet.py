@@ -32,6 +32,10 @@ def main():
     p.add_argument("--stock", action="store_true",
                    help="stock torch SyncBatchNorm+DDP comparison line "
                         "(identical architecture and init)")
+    p.add_argument("--graph", action="store_true",
+                   help="capture the whole train step in one hipGraph (msbn "
+                        "only): the detection regime is host/launch-bound — "
+                        "~70 small BN layers per step")
     p.add_argument("--channels-last", action="store_true",
                    help="channels_last memory format (NCHW is the measured "
                         "best for detection shapes on this stack, and stock "
@@ -54,6 +58,15 @@ def main():
                                 rank=rank)
 
     torch.manual_seed(11)
+    import contextlib
+
+    use_graph = args.graph and use_cuda and not args.stock
+    # NOTE: warmup AND capture run on `side`; grad accumulators + optimizer
+    # state are created during warmup, so they live on the capture stream
+    # (the DDP-under-graph recipe, tools/nccl_world1_check.py).  At world>1
+    # wrap DDP construction in torch.cuda.stream(side) as bench.py does.
+    side = torch.cuda.Stream() if use_graph else None
+    del contextlib
     if args.stock:
         from msbn.models import convert_to_torch_batchnorm
 
@@ -101,15 +114,37 @@ def main():
         opt.step()
         return loss
 
-    for _ in range(args.warmup):
-        step()
+    graph = None
+    if use_graph:
+        with torch.cuda.stream(side):
+            for _ in range(max(args.warmup, 3)):
+                step()
+        torch.cuda.synchronize()
+        try:
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph, stream=side):
+                step()
+            for _ in range(args.warmup):
+                graph.replay()
+        except Exception as e:
+            print(f"[bench_retinanet] hipGraph capture failed ({e!r}); eager",
+                  file=sys.stderr)
+            graph = None
+            torch.cuda.synchronize()
+    else:
+        for _ in range(args.warmup):
+            step()
     if world > 1:
         dist.barrier()
     if use_cuda:
         torch.cuda.synchronize()
     t0 = time.perf_counter()
-    for _ in range(args.steps):
-        step()
+    if graph is not None:
+        for _ in range(args.steps):
+            graph.replay()
+    else:
+        for _ in range(args.steps):
+            step()
     if use_cuda:
         torch.cuda.synchronize()
     if world > 1:
@@ -138,6 +173,7 @@ def main():
                        "image": f"3x{H}x{W}",
                        "per_gpu_batch": args.batch_size,
                        "parallelism": f"dp{world}",
+                       "hip_graph": graph is not None,
                        "memory_format": "channels_last" if args.channels_last
                        else "contiguous",
                        "impl": "stock" if args.stock else "msbn"},
