@@ -119,7 +119,11 @@ def compute_sync_stats(
             (world_size, 2 * C + 1), dtype=torch.float32, device=input.device
         )
         from msbn.utils.logging import comm_log
+        from msbn.utils import debug as _dbg
 
+        if _dbg.enabled():
+            _dbg.verify_collective("syncbn.fwd.all_gather", packed,
+                                   process_group)
         if _is_nccl_like(process_group):
             dist.all_gather_into_tensor(packed_all, packed, group=process_group)
         else:
@@ -269,6 +273,10 @@ class SyncBatchNormFunction(torch.autograd.Function):
             if need_input_g:
                 if getattr(ctx, "use_sync", world_size > 1):
                     combined, copied = _combined_view(sum_dy, sum_dy_xmu, C)
+                    from msbn.utils import debug as _dbg
+                    if _dbg.enabled():
+                        _dbg.verify_collective("syncbn.bwd.all_reduce",
+                                               combined, process_group)
                     dist.all_reduce(
                         combined, dist.ReduceOp.SUM, group=process_group
                     )

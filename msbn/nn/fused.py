@@ -169,6 +169,10 @@ class SyncBatchNormActFunction(torch.autograd.Function):
         if need_input_g or need_res_g:
             if world_size > 1:
                 combined, copied = _combined_view(sum_dy, sum_dy_xmu, C)
+                from msbn.utils import debug as _dbg
+                if _dbg.enabled():
+                    _dbg.verify_collective("syncbn.bwd.all_reduce",
+                                           combined, process_group)
                 dist.all_reduce(combined, dist.ReduceOp.SUM, group=process_group)
                 from msbn.utils.logging import comm_log
                 comm_log.record("all_reduce", combined.numel() * 4,

@@ -103,6 +103,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("set_comm_dtype", &msbn::Reducer::set_comm_dtype)
       .def("set_nan_check", &msbn::Reducer::set_nan_check)
       .def("set_div_factor", &msbn::Reducer::set_div_factor)
+      .def("set_python_comm_hook", &msbn::Reducer::set_python_comm_hook,
+           py::arg("state"), py::arg("hook"), py::arg("bucket_cls"))
       .def("div_factor", &msbn::Reducer::div_factor)
       .def("find_unused", &msbn::Reducer::find_unused, py::arg("outputs"))
       .def("finalize_backward", &msbn::Reducer::finalize_backward,

@@ -280,6 +280,21 @@ class Reducer : public std::enable_shared_from_this<Reducer> {
     div_factor_ = f;
   }
 
+  // Arbitrary Python comm hook (stock register_comm_hook's general form).
+  // The hook REPLACES the built-in allreduce+divide: it receives
+  // (state, GradBucket) and returns a torch.futures.Future resolving to the
+  // reduced flat tensor.  Python hooks run at FINALIZE, in bucket order, on
+  // the engine-callback thread (documented caveat: no backward overlap —
+  // the C++ fast path with the builtin hooks keeps the overlap).
+  void set_python_comm_hook(py::object state, py::object hook,
+                            py::object bucket_cls) {
+    std::lock_guard<std::mutex> lock(mutex_);
+    py_state_ = std::move(state);
+    py_hook_ = std::move(hook);
+    py_bucket_cls_ = std::move(bucket_cls);
+    has_py_hook_ = !py_hook_.is_none();
+  }
+
   double div_factor() const { return div_factor_; }
 
   // Walk the autograd graph backward from `outputs` on the C++ side (no
@@ -347,22 +362,30 @@ class Reducer : public std::enable_shared_from_this<Reducer> {
   }
 
   void finalize_backward() {
-    std::lock_guard<std::mutex> lock(mutex_);
-    expect_autograd_hooks_ = false;
-    // every bucket must have launched
-    if (next_bucket_ != buckets_.size()) {
-      std::ostringstream oss;
-      oss << "msbn DDP: expected all gradient buckets to be ready at the end "
-             "of backward ("
-          << next_bucket_ << "/" << buckets_.size()
-          << " launched). Some parameters did not receive gradients; unused "
-             "parameter indices: [";
-      for (size_t i = 0; i < params_.size(); ++i)
-        if (!param_ready_[i]) oss << i << ", ";
-      oss << "]. Run the DDP wrapper with find_unused_parameters=True.";
-      finalize_queued_ = false;
-      TORCH_CHECK(false, oss.str());
+    {
+      std::lock_guard<std::mutex> lock(mutex_);
+      expect_autograd_hooks_ = false;
+      // every bucket must have launched
+      if (next_bucket_ != buckets_.size()) {
+        std::ostringstream oss;
+        oss << "msbn DDP: expected all gradient buckets to be ready at the "
+               "end of backward ("
+            << next_bucket_ << "/" << buckets_.size()
+            << " launched). Some parameters did not receive gradients; unused "
+               "parameter indices: [";
+        for (size_t i = 0; i < params_.size(); ++i)
+          if (!param_ready_[i]) oss << i << ", ";
+        oss << "]. Run the DDP wrapper with find_unused_parameters=True.";
+        finalize_queued_ = false;
+        TORCH_CHECK(false, oss.str());
+      }
     }
+    // Python comm hooks run here, OUTSIDE the reducer mutex (GIL + mutex
+    // inversion with other Python threads otherwise) — launch all, then
+    // wait in bucket order.  No hooks can fire concurrently:
+    // expect_autograd_hooks_ is already false.
+    if (has_py_hook_) run_python_hooks();
+    std::lock_guard<std::mutex> lock(mutex_);
     for (auto& b : buckets_) {
       if (b.work) {
         b.work->wait();
@@ -555,6 +578,8 @@ class Reducer : public std::enable_shared_from_this<Reducer> {
                   "msbn Reducer: non-finite gradient detected in bucket "
                   "(params ", b.param_indices, ") before all-reduce");
     }
+    // a Python comm hook REPLACES the allreduce+divide; it runs at finalize
+    if (has_py_hook_) return;
     if (div_factor_ != 1.0) b.flat.div_(div_factor_);
     c10d::AllreduceOptions opts;
     if (comm_dtype_.has_value() &&
@@ -566,6 +591,30 @@ class Reducer : public std::enable_shared_from_this<Reducer> {
       b.wire = at::Tensor();
       std::vector<at::Tensor> v{b.flat};
       b.work = pg_->allreduce(v, opts);
+    }
+  }
+
+  void run_python_hooks() {
+    py::gil_scoped_acquire gil;
+    std::vector<py::object> futs;
+    futs.reserve(buckets_.size());
+    for (size_t i = 0; i < buckets_.size(); ++i) {
+      auto& b = buckets_[i];
+      py::list views;
+      for (auto& v : b.views) views.append(v);
+      py::object bucket =
+          py_bucket_cls_(b.flat, views, (int64_t)i,
+                         i + 1 == buckets_.size());
+      futs.push_back(py_hook_(py_state_, bucket));
+    }
+    for (size_t i = 0; i < buckets_.size(); ++i) {
+      py::object res = futs[i].attr("wait")();
+      auto out = res.cast<at::Tensor>();
+      auto& flat = buckets_[i].flat;
+      if (out.data_ptr() != flat.data_ptr()) {
+        flat.copy_(out.reshape({-1}).to(flat.scalar_type()),
+                   /*non_blocking=*/true);
+      }
     }
   }
 
@@ -584,6 +633,10 @@ class Reducer : public std::enable_shared_from_this<Reducer> {
 
   c10::optional<at::ScalarType> comm_dtype_;
   bool nan_check_ = false;
+  bool has_py_hook_ = false;
+  py::object py_state_;
+  py::object py_hook_;
+  py::object py_bucket_cls_;
 
   std::vector<int64_t> last_ready_ns_;   // steady_clock ns per param
   int64_t first_ready_ns_ = 0;

@@ -489,12 +489,19 @@ class DistributedDataParallel(Module):
                 torch.autograd.backward([o.sum() for o in outs])
 
     def register_comm_hook(self, state, hook):
-        """Gradient-communication hook (stock register_comm_hook).  The three
-        builtin c10d hooks are recognized BY FUNCTION IDENTITY (not name
-        sniffing) and map onto the C++ reducer: allreduce_hook is the default
-        path, fp16/bf16 compress hooks become a wire-dtype cast.  Arbitrary
-        Python hooks cannot run inside the C++ backward path and are rejected
-        explicitly; use set_comm_dtype for custom compression dtypes."""
+        """Gradient-communication hook (stock register_comm_hook).
+
+        The three builtin c10d hooks are recognized BY FUNCTION IDENTITY
+        (not name sniffing) and map onto the C++ reducer's fast path:
+        allreduce_hook is the default, fp16/bf16 compress hooks become a
+        wire-dtype cast — full backward/comm overlap is preserved.
+
+        Any OTHER callable runs as a real Python hook: the reducer calls
+        ``hook(state, GradBucket)`` per bucket at finalize (launched in
+        bucket order, futures awaited in order).  The hook replaces the
+        all-reduce AND the world-size division, matching stock semantics.
+        Caveat (documented): Python hooks run after backward compute — no
+        overlap; prefer the builtins for production."""
         try:
             from torch.distributed.algorithms.ddp_comm_hooks import (
                 default_hooks as _dh,
@@ -504,21 +511,21 @@ class DistributedDataParallel(Module):
         if _dh is not None:
             if hook is _dh.allreduce_hook:
                 self.reducer.set_comm_dtype(None)
+                self.reducer.set_python_comm_hook(None, None, None)
                 return
             if hook is _dh.fp16_compress_hook:
                 self.reducer.set_comm_dtype(torch.float16)
+                self.reducer.set_python_comm_hook(None, None, None)
                 return
             if hook is _dh.bf16_compress_hook:
                 self.reducer.set_comm_dtype(torch.bfloat16)
+                self.reducer.set_python_comm_hook(None, None, None)
                 return
-        raise NotImplementedError(
-            "msbn DDP supports the builtin c10d comm hooks (allreduce_hook, "
-            "fp16_compress_hook, bf16_compress_hook from torch.distributed."
-            "algorithms.ddp_comm_hooks.default_hooks), matched by identity; "
-            f"got {getattr(hook, '__name__', repr(hook))!r}. Custom Python "
-            "comm hooks are not run inside the C++ backward path — use "
-            "set_comm_dtype() for custom wire dtypes."
-        )
+        if not callable(hook):
+            raise TypeError(f"comm hook must be callable, got {hook!r}")
+        from msbn.parallel.comm_hooks import GradBucket
+
+        self.reducer.set_python_comm_hook(state, hook, GradBucket)
 
     def set_comm_dtype(self, dtype):
         """Cast gradient buckets to `dtype` for the wire (None to disable)."""

@@ -168,26 +168,64 @@ def _comm_hook_identity_body(rank, world):
     from torch.distributed.algorithms.ddp_comm_hooks import default_hooks as dh
 
     net = msbn.parallel.DistributedDataParallel(torch.nn.Linear(8, 8))
-    # builtins accepted by identity
+    # builtins accepted by identity -> C++ fast path (no python hook set)
     net.register_comm_hook(None, dh.fp16_compress_hook)
     net.register_comm_hook(None, dh.bf16_compress_hook)
     net.register_comm_hook(None, dh.allreduce_hook)
-
-    # a lookalike name must NOT silently enable compression
-    def my_bf16_logging_hook(state, bucket):  # pragma: no cover - never runs
-        return dh.allreduce_hook(state, bucket)
-
+    # non-callables rejected
     try:
-        net.register_comm_hook(None, my_bf16_logging_hook)
-        raise AssertionError("lookalike hook was accepted")
-    except NotImplementedError:
+        net.register_comm_hook(None, "not-a-hook")
+        raise AssertionError("non-callable hook was accepted")
+    except TypeError:
         pass
-    # one step still works with the default hook restored
+    # one step works with the default hook
     net(torch.randn(4, 8)).sum().backward()
 
 
 def test_comm_hook_identity(tmp_path):
     _spawn("_comm_hook_identity_body", tmp_path, world=2)
+
+
+def _python_comm_hook_body(rank, world):
+    """Arbitrary Python comm hooks execute for real: a hook that all-reduces
+    and then DOUBLES the bucket must yield grads == 2 x the default path's.
+    A lookalike-named hook no longer silently enables compression — it runs
+    as written."""
+    import msbn
+
+    torch.manual_seed(9)
+    x = torch.randn(4, 8, generator=torch.Generator().manual_seed(50 + rank))
+
+    def grads_with(hook):
+        torch.manual_seed(9)
+        net = msbn.parallel.DistributedDataParallel(torch.nn.Linear(8, 8))
+        if hook is not None:
+            net.register_comm_hook(None, hook)
+        net(x).pow(2).sum().backward()
+        return torch.cat([p.grad.flatten() for p in net.module.parameters()])
+
+    calls = []
+
+    def my_bf16_logging_hook(state, bucket):
+        # name contains "bf16" but this hook does allreduce x2, not
+        # compression — it must run AS WRITTEN (identity matching)
+        calls.append(bucket.index())
+        t = bucket.buffer()
+        t.div_(world)
+        work = dist.all_reduce(t, group=None, async_op=True)
+        fut = work.get_future()
+        return fut.then(lambda f: f.value()[0] * 2.0)
+
+    base = grads_with(None)
+    doubled = grads_with(my_bf16_logging_hook)
+    assert calls, "python hook never ran"
+    assert torch.allclose(doubled, 2.0 * base, atol=1e-5), (
+        doubled / base,
+    )
+
+
+def test_python_comm_hook(tmp_path):
+    _spawn("_python_comm_hook_body", tmp_path, world=2)
 
 
 # ---------------------------------------------- verify_params mismatch raise
@@ -409,6 +447,52 @@ def test_find_unused_cpp_matches_python(tmp_path):
             assert cpp_unused == py_unused, (use_b, cpp_unused, py_unused)
     finally:
         dist.destroy_process_group()
+
+
+# ------------------------------------------- collective-agreement debugging
+def _debug_collectives_ok_body(rank, world):
+    """MSBN_DEBUG_COLLECTIVES=1: matched collectives pass the shadow-group
+    verification (stock ProcessGroupWrapper under DEBUG=DETAIL, §5.2)."""
+    import msbn
+
+    os.environ["MSBN_DEBUG_COLLECTIVES"] = "1"
+    try:
+        torch.manual_seed(2)
+        bn = msbn.nn.SyncBatchNorm(4)
+        bn.train()
+        x = torch.randn(3, 4, 5, 5, requires_grad=True)
+        y = bn(x)
+        y.sum().backward()
+        assert x.grad is not None
+    finally:
+        os.environ.pop("MSBN_DEBUG_COLLECTIVES", None)
+
+
+def test_debug_collectives_matched(tmp_path):
+    _spawn("_debug_collectives_ok_body", tmp_path, world=2)
+
+
+def _debug_collectives_mismatch_body(rank, world):
+    """Desynchronized ranks (different channel counts -> different message
+    sizes) raise a clean error on EVERY rank instead of hanging."""
+    import msbn
+
+    os.environ["MSBN_DEBUG_COLLECTIVES"] = "1"
+    try:
+        C = 4 if rank == 0 else 6
+        bn = msbn.nn.SyncBatchNorm(C)
+        bn.train()
+        try:
+            bn(torch.randn(3, C, 5, 5))
+            raise AssertionError(f"rank {rank}: desync not detected")
+        except RuntimeError as e:
+            assert "collective-agreement" in str(e), str(e)
+    finally:
+        os.environ.pop("MSBN_DEBUG_COLLECTIVES", None)
+
+
+def test_debug_collectives_mismatch(tmp_path):
+    _spawn("_debug_collectives_mismatch_body", tmp_path, world=2)
 
 
 # ----------------------------------------------------------- backward stats
