@@ -57,3 +57,9 @@ def test_nccl_world1_input_h2d_and_backward_stats():
     """CPU inputs moved to GPU on a side stream in DDP.forward; per-param
     grad-ready timestamps surfaced via _get_ddp_logging_data."""
     _run_case("h2d")
+
+
+def test_nccl_world1_allocator_stress():
+    """Async bucket all-reduces stay correct under caching-allocator churn:
+    30 DDP steps with random temporaries in flight track a plain clone."""
+    _run_case("stress")

@@ -170,6 +170,46 @@ def case_h2d():
     print("CASE h2d OK")
 
 
+def case_stress():
+    """Allocator-safety of the async bucket path under churn: every
+    iteration allocates/frees random-size temporaries while the reducer's
+    bucket all-reduces are in flight on PGNCCL's comm stream.  30 DDP steps
+    must track a plain (no-DDP) clone bit-for-bit at world 1 — any
+    allocator reuse of a bucket block mid-flight corrupts grads and
+    diverges the models (SURVEY.md §5.2 recordStream/stash problem)."""
+    import msbn
+
+    torch.manual_seed(31)
+    gen = torch.Generator().manual_seed(77)
+    model = make_model(seed=31)
+    ddp = msbn.parallel.DistributedDataParallel(model, device_ids=[0],
+                                                output_device=0)
+    ref = make_model(seed=31)  # identical init
+    opt = torch.optim.SGD(ddp.parameters(), lr=0.03, momentum=0.9)
+    ropt = torch.optim.SGD(ref.parameters(), lr=0.03, momentum=0.9)
+    loss_fn = torch.nn.CrossEntropyLoss()
+    junk = []
+    for it in range(30):
+        x = torch.randn(8, 3, 8, 8, generator=gen).cuda()
+        y = torch.randint(0, 10, (8,), generator=gen).cuda()
+        opt.zero_grad(set_to_none=True)
+        loss_fn(ddp(x), y).backward()
+        # allocator churn while comm-stream work may still be in flight
+        junk.clear()
+        for _ in range(6):
+            n = int(torch.randint(1, 4 << 20, (1,), generator=gen))
+            junk.append(torch.empty(n, device="cuda").fill_(float(it)))
+        opt.step()
+        ropt.zero_grad(set_to_none=True)
+        loss_fn(ref(x), y).backward()
+        ropt.step()
+    torch.cuda.synchronize()
+    for (n, p), (_, rp) in zip(ddp.module.named_parameters(),
+                               ref.named_parameters()):
+        torch.testing.assert_close(p, rp, atol=1e-5, rtol=1e-5, msg=n)
+    print("CASE stress OK")
+
+
 def main():
     case = sys.argv[1] if len(sys.argv) > 1 else "ddp"
     torch.cuda.set_device(0)
@@ -181,6 +221,8 @@ def main():
             case_graph()
         elif case == "h2d":
             case_h2d()
+        elif case == "stress":
+            case_stress()
         else:
             raise SystemExit(f"unknown case {case}")
     finally:
