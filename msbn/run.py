@@ -100,6 +100,10 @@ def _worker_env(args, local_rank: int, nproc: int, port: int, restart: int):
         OMP_NUM_THREADS=env.get("OMP_NUM_THREADS", "1"),
     )
     env.setdefault("TORCH_NCCL_ASYNC_ERROR_HANDLING", "1")
+    if getattr(args, "_agent_store_hosted", False):
+        # workers attach to the launcher-hosted TCPStore instead of rank 0
+        # hosting one (stock rendezvous.py:160 contract)
+        env["TORCHELASTIC_USE_AGENT_STORE"] = str(True)
     return env
 
 
@@ -141,6 +145,19 @@ def _kill_group(procs):
             p.kill()
 
 
+def _host_agent_store(addr: str, port: int):
+    """Host the rendezvous TCPStore in the LAUNCHER (stock elastic-agent
+    behavior, local_elastic_agent.py + TORCHELASTIC_USE_AGENT_STORE): the
+    store outlives worker restarts, so re-spawned groups rendezvous on the
+    same port with no rank-0 hosting race."""
+    try:
+        from torch.distributed import TCPStore
+
+        return TCPStore(addr, port, is_master=True, multi_tenant=True)
+    except Exception:
+        return None  # workers fall back to rank-0-hosted store
+
+
 def run(args) -> int:
     if getattr(args, "rdzv_endpoint", None):
         host, _, port = args.rdzv_endpoint.partition(":")
@@ -156,16 +173,30 @@ def run(args) -> int:
         # disagree and rendezvous could never form.  Match the stock
         # launchers' fixed default.
         args.master_port = 29500
-    nproc = (
-        _device_count()
-        if str(args.nproc_per_node) in ("auto", "gpu")
-        else int(args.nproc_per_node)
-    )
+    auto_nproc = str(args.nproc_per_node) in ("auto", "gpu")
+    nproc = _device_count() if auto_nproc else int(args.nproc_per_node)
     if args.run_id is None:
         args.run_id = uuid.uuid4().hex[:8]
     restarts = 0
+    # one port + one agent-hosted store for the whole (restartable) job
+    fixed_port = args.master_port or _free_port()
+    agent_store = None
+    if args.nnodes == 1 or args.node_rank == 0:
+        agent_store = _host_agent_store(args.master_addr, fixed_port)
+    args._agent_store_hosted = agent_store is not None  # ref keeps the store alive
     while True:
-        port = args.master_port or _free_port()
+        port = fixed_port
+        if auto_nproc and restarts > 0:
+            # elastic-style scale-down: if a device died with the failed
+            # group, respawn with the surviving count
+            new_nproc = _device_count()
+            if new_nproc != nproc:
+                print(
+                    f"[msbn.run] device count changed {nproc} -> {new_nproc}; "
+                    "re-forming group at the new size",
+                    file=sys.stderr,
+                )
+                nproc = new_nproc
         procs = _spawn_group(args, nproc, port, restarts)
         failed_rc = None
         live = list(procs)
