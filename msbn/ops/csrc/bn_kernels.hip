@@ -43,28 +43,33 @@ __global__ void bn_stats_partial_nchw(const T* __restrict__ x,
   const int64_t n1 = i64min(n0 + chunkN, N);
   const int64_t s0 = blockIdx.z * chunkS;
   const int64_t s1 = i64min(s0 + chunkS, S);
-  double a = 0.0, b = 0.0;
+  AccPair acc;
+  int since_flush = 0;
   for (int64_t n = n0; n < n1; ++n) {
     const T* row = x + (n * C + c) * S;
     if (V == 1) {
       for (int64_t s = s0 + threadIdx.x; s < s1; s += blockDim.x) {
-        float v = to_f(nt_load1(&row[s]));
-        a += v;
-        b += (double)v * v;
+        acc.add(to_f(nt_load1(&row[s])));
+        if (++since_flush == MSBN_ACC_FLUSH) {
+          acc.flush();
+          since_flush = 0;
+        }
       }
     } else {
       for (int64_t s = s0 + (int64_t)threadIdx.x * V; s < s1;
            s += (int64_t)blockDim.x * V) {
         Pack<T, V> pk = nt_load<T, V>(&row[s]);
 #pragma unroll
-        for (int k = 0; k < V; ++k) {
-          float v = to_f(pk.v[k]);
-          a += v;
-          b += (double)v * v;
+        for (int k = 0; k < V; ++k) acc.add(to_f(pk.v[k]));
+        if (++since_flush == MSBN_ACC_FLUSH / V) {
+          acc.flush();
+          since_flush = 0;
         }
       }
     }
   }
+  acc.flush();
+  double a = acc.a, b = acc.b;
   __shared__ double lds[2 * (MSBN_BLOCK / MSBN_WAVE)];
   block_reduce_pair(a, b, lds);
   if (threadIdx.x == 0) {
@@ -87,13 +92,18 @@ __global__ void bn_stats_partial_nchw_flat(const T* __restrict__ x,
   const int64_t NS = N * S;
   const int64_t p0 = (int64_t)blockIdx.y * chunk_len;
   const int64_t p1 = i64min(p0 + chunk_len, NS);
-  double a = 0.0, b = 0.0;
+  AccPair acc;
+  int since_flush = 0;
   for (int64_t p = p0 + threadIdx.x; p < p1; p += blockDim.x) {
     const int64_t n = p / S, s = p - n * S;
-    const float v = to_f(nt_load1(&x[(n * C + c) * S + s]));
-    a += v;
-    b += (double)v * v;
+    acc.add(to_f(nt_load1(&x[(n * C + c) * S + s])));
+    if (++since_flush == MSBN_ACC_FLUSH) {
+      acc.flush();
+      since_flush = 0;
+    }
   }
+  acc.flush();
+  double a = acc.a, b = acc.b;
   __shared__ double lds[2 * (MSBN_BLOCK / MSBN_WAVE)];
   block_reduce_pair(a, b, lds);
   if (threadIdx.x == 0) {
@@ -119,22 +129,29 @@ __global__ void bn_stats_partial_nhwc(const T* __restrict__ x,
   const int64_t c = (int64_t)blockIdx.x * lpr * V + (int64_t)lane * V;
   const bool inb = active && (c + V <= C);
 
-  double a[V], b[V];
-#pragma unroll
-  for (int k = 0; k < V; ++k) a[k] = b[k] = 0.0;
+  AccPair acc[V];
 
   if (inb) {
     const int64_t r0 = (int64_t)blockIdx.y * chunk_rows;
     const int64_t r1 = i64min(r0 + chunk_rows, rows);
+    int since_flush = 0;
     for (int64_t r = r0 + rowoff; r < r1; r += rpi) {
       Pack<T, V> pk = nt_load<T, V>(&x[r * C + c]);
 #pragma unroll
-      for (int k = 0; k < V; ++k) {
-        float v = to_f(pk.v[k]);
-        a[k] += v;
-        b[k] += (double)v * v;
+      for (int k = 0; k < V; ++k) acc[k].add(to_f(pk.v[k]));
+      if (++since_flush == MSBN_ACC_FLUSH) {
+#pragma unroll
+        for (int k = 0; k < V; ++k) acc[k].flush();
+        since_flush = 0;
       }
     }
+  }
+  double a[V], b[V];
+#pragma unroll
+  for (int k = 0; k < V; ++k) {
+    acc[k].flush();
+    a[k] = acc[k].a;
+    b[k] = acc[k].b;
   }
 
   // LDS tree-reduce across the rowoff dimension (generic, non-pow2 safe).
@@ -423,7 +440,8 @@ __global__ void bn_bwd_reduce_partial_nchw(const T* __restrict__ dy,
   const int64_t n1 = i64min(n0 + chunkN, N);
   const int64_t s0 = blockIdx.z * chunkS;
   const int64_t s1 = i64min(s0 + chunkS, S);
-  double a = 0.0, b = 0.0;
+  AccPair acc;
+  int since_flush = 0;
   for (int64_t n = n0; n < n1; ++n) {
     const int64_t base = (n * C + c) * S;
     if (V == 1) {
@@ -436,8 +454,11 @@ __global__ void bn_bwd_reduce_partial_nchw(const T* __restrict__ dy,
           if (z <= 0.f) g = 0.f;
         }
         if (GMOUT) nt_store1(&gm_out[base + s], from_f<T>(g));
-        a += g;
-        b += (double)g * (xv - m);
+        acc.add2(g, g * (xv - m));
+        if (++since_flush == MSBN_ACC_FLUSH) {
+          acc.flush();
+          since_flush = 0;
+        }
       }
     } else {
       for (int64_t s = s0 + (int64_t)threadIdx.x * V; s < s1;
@@ -456,13 +477,18 @@ __global__ void bn_bwd_reduce_partial_nchw(const T* __restrict__ dy,
             if (z <= 0.f) g = 0.f;
           }
           if (GMOUT) pm.v[k] = from_f<T>(g);
-          a += g;
-          b += (double)g * (xv - m);
+          acc.add2(g, g * (xv - m));
         }
         if (GMOUT) nt_store<T, V>(&gm_out[base + s], pm);
+        if (++since_flush == MSBN_ACC_FLUSH / V) {
+          acc.flush();
+          since_flush = 0;
+        }
       }
     }
   }
+  acc.flush();
+  double a = acc.a, b = acc.b;
   __shared__ double lds[2 * (MSBN_BLOCK / MSBN_WAVE)];
   block_reduce_pair(a, b, lds);
   if (threadIdx.x == 0) {
@@ -488,7 +514,8 @@ __global__ void bn_bwd_reduce_partial_nchw_flat(
   const int64_t NS = N * S;
   const int64_t p0 = (int64_t)blockIdx.y * chunk_len;
   const int64_t p1 = i64min(p0 + chunk_len, NS);
-  double a = 0.0, b = 0.0;
+  AccPair acc;
+  int since_flush = 0;
   for (int64_t p = p0 + threadIdx.x; p < p1; p += blockDim.x) {
     const int64_t n = p / S, s = p - n * S;
     const int64_t e = (n * C + c) * S + s;
@@ -500,9 +527,14 @@ __global__ void bn_bwd_reduce_partial_nchw_flat(
       if (z <= 0.f) g = 0.f;
     }
     if (GMOUT) nt_store1(&gm_out[e], from_f<T>(g));
-    a += g;
-    b += (double)g * (xv - m);
+    acc.add2(g, g * (xv - m));
+    if (++since_flush == MSBN_ACC_FLUSH) {
+      acc.flush();
+      since_flush = 0;
+    }
   }
+  acc.flush();
+  double a = acc.a, b = acc.b;
   __shared__ double lds[2 * (MSBN_BLOCK / MSBN_WAVE)];
   block_reduce_pair(a, b, lds);
   if (threadIdx.x == 0) {
@@ -530,13 +562,10 @@ __global__ void bn_bwd_reduce_partial_nhwc(const T* __restrict__ dy,
   const int64_t c = (int64_t)blockIdx.x * lpr * V + (int64_t)lane * V;
   const bool inb = active && (c + V <= C);
 
-  double a[V], b[V];
+  AccPair acc[V];
   float m[V], scv[V], shv[V];
 #pragma unroll
-  for (int k = 0; k < V; ++k) {
-    a[k] = b[k] = 0.0;
-    m[k] = scv[k] = shv[k] = 0.f;
-  }
+  for (int k = 0; k < V; ++k) m[k] = scv[k] = shv[k] = 0.f;
   if (inb) {
 #pragma unroll
     for (int k = 0; k < V; ++k) {
@@ -548,6 +577,7 @@ __global__ void bn_bwd_reduce_partial_nhwc(const T* __restrict__ dy,
     }
     const int64_t r0 = (int64_t)blockIdx.y * chunk_rows;
     const int64_t r1 = i64min(r0 + chunk_rows, rows);
+    int since_flush = 0;
     for (int64_t r = r0 + rowoff; r < r1; r += rpi) {
       Pack<T, V> pg = nt_load<T, V>(&dy[r * C + c]);
       Pack<T, V> px = nt_load<T, V>(&x[r * C + c]);
@@ -563,11 +593,22 @@ __global__ void bn_bwd_reduce_partial_nhwc(const T* __restrict__ dy,
           if (z <= 0.f) g = 0.f;
         }
         if (GMOUT) pm.v[k] = from_f<T>(g);
-        a[k] += g;
-        b[k] += (double)g * (xv - m[k]);
+        acc[k].add2(g, g * (xv - m[k]));
       }
       if (GMOUT) nt_store<T, V>(&gm_out[r * C + c], pm);
+      if (++since_flush == MSBN_ACC_FLUSH) {
+#pragma unroll
+        for (int k = 0; k < V; ++k) acc[k].flush();
+        since_flush = 0;
+      }
     }
+  }
+  double a[V], b[V];
+#pragma unroll
+  for (int k = 0; k < V; ++k) {
+    acc[k].flush();
+    a[k] = acc[k].a;
+    b[k] = acc[k].b;
   }
   // LDS layout [k][tid] with separate sum/sumsq planes: lane l of a wave
   // lands on bank (2*l) mod 64 for ds_*_b64 -> conflict-free (the [tid][k]

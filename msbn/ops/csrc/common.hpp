@@ -150,4 +150,34 @@ __device__ __forceinline__ int64_t i64min(int64_t a, int64_t b) {
   return a < b ? a : b;
 }
 
+// -------------------------------------------- chunked fp32->fp64 accumulator
+// The streaming reduction loops are VALU-bound on MI355X when they
+// accumulate every element in fp64 (f64 vector ops run at half rate: 16
+// half-rate f64 ops per 16 B load caps the stats kernels at ~4.8 TB/s of
+// the 6.3 TB/s ceiling).  Accumulate a short fp32 chunk (<= kAccFlush loop
+// iterations) and FLUSH into the fp64 totals periodically: full-rate inner
+// loop, and the fp64 totals still kill E[x^2]-E[x]^2 cancellation.  fp32
+// chunk error is ~sqrt(n)*2^-24 over <=512 elements — far below the fp32
+// reference tolerance the tests compare against.
+#define MSBN_ACC_FLUSH 64
+
+struct AccPair {
+  double a = 0.0, b = 0.0;  // fp64 totals
+  float fa = 0.f, fb = 0.f;  // fp32 chunk
+
+  __device__ __forceinline__ void add(float v) {
+    fa += v;
+    fb += v * v;
+  }
+  __device__ __forceinline__ void add2(float s, float sq) {
+    fa += s;
+    fb += sq;
+  }
+  __device__ __forceinline__ void flush() {
+    a += (double)fa;
+    b += (double)fb;
+    fa = fb = 0.f;
+  }
+};
+
 }  // namespace msbn
