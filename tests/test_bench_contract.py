@@ -70,12 +70,17 @@ def test_bench_multirank_cpu_gloo(tmp_path):
     import subprocess
     import sys
 
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
     env = dict(os.environ)
     env["PYTHONPATH"] = REPO
     r = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", "29617", os.path.join(REPO, "bench.py"),
+         "--master-port", str(port), os.path.join(REPO, "bench.py"),
          "--gpus", "2", "--steps", "1", "--warmup", "0",
          "--batch-size", "2", "--model", "resnet18", "--dtype", "fp32",
          "--memory-format", "contiguous"],
