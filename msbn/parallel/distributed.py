@@ -259,15 +259,29 @@ class DistributedDataParallel(Module):
         return unused
 
     def _notify_join_context(self):
-        """First collective of a joined-training iteration: count of ranks
-        still feeding real data (stock Join.notify_join_context, S11).
-        Returns that count, or None when no join() context is active."""
+        """First collective of a joined-training iteration (stock
+        Join.notify_join_context + _check_global_requires_backward_grad_sync,
+        S11): ONE 2-element all_reduce carrying [I-have-data,
+        I-will-sync-grads].  Shadowing ranks contribute [0, 0]; trainers
+        under no_sync() contribute [1, 0] so shadows skip their bucket
+        all-reduces too.  Mixed sync settings among trainers raise on every
+        rank.  Returns the active-trainer count, or None outside join()."""
         cfg = self._join_config
         if cfg is None or cfg.get("shadowing"):
             return None
-        t = torch.ones(1, dtype=torch.float32, device=self._comm_device())
+        t = torch.tensor(
+            [1.0, 1.0 if self.require_backward_grad_sync else 0.0],
+            dtype=torch.float32, device=self._comm_device(),
+        )
         dist.all_reduce(t, op=dist.ReduceOp.SUM, group=self.process_group)
-        active = int(t.item())
+        active = int(t[0].item())
+        wants_sync = int(t[1].item())
+        if 0 < wants_sync < active:
+            raise RuntimeError(
+                "msbn DDP.join(): require_backward_grad_sync differs across "
+                f"active ranks ({wants_sync}/{active} syncing) — no_sync() "
+                "must be entered by every rank on the same iterations"
+            )
         world = dist.get_world_size(self.process_group)
         if cfg["throw_on_early_termination"] and active < world:
             raise RuntimeError(
@@ -338,12 +352,14 @@ class DistributedDataParallel(Module):
         with torch.autograd.profiler.record_function(
             "msbn.DistributedDataParallel.forward"
         ):
-            if torch.is_grad_enabled() and self.require_backward_grad_sync:
+            if torch.is_grad_enabled() and self._join_config is not None:
+                # notify runs even under no_sync: the agreement flag tells
+                # shadowing ranks whether bucket all-reduces happen this
+                # iteration (stock S11 semantics)
                 self._notify_join_context()
-                if self._join_config is not None and not self._join_config.get(
-                    "shadowing"
-                ):
+                if not self._join_config.get("shadowing"):
                     self._record_join_input_spec(inputs, kwargs)
+            if torch.is_grad_enabled() and self.require_backward_grad_sync:
                 # one-shot arrival-order bucket rebuild (stock _rebuild_buckets).
                 # Gated off under find_unused without static_graph: per-rank
                 # arrival completeness can differ across ranks there, and a
@@ -439,9 +455,10 @@ class DistributedDataParallel(Module):
         cfg = self._join_config
         device = self._comm_device()
         while True:
-            t = torch.zeros(1, dtype=torch.float32, device=device)
+            t = torch.zeros(2, dtype=torch.float32, device=device)
             dist.all_reduce(t, op=dist.ReduceOp.SUM, group=self.process_group)
-            active = int(t.item())
+            active = int(t[0].item())
+            wants_sync = int(t[1].item())
             if active == 0:
                 # All ranks are done.  Joined ranks never ran the optimizer
                 # during their shadow steps, so their parameters are stale by
@@ -480,13 +497,21 @@ class DistributedDataParallel(Module):
                 p.grad = None
             inputs, kwargs = self._join_input_spec
             cfg["shadowing"] = True
+            # mirror the trainers' no_sync state: when they skip gradient
+            # sync this iteration (wants_sync == 0), the shadow must not
+            # post bucket all-reduces or enter the rebuild broadcast either
+            old_sync = self.require_backward_grad_sync
+            self.require_backward_grad_sync = wants_sync > 0
             try:
                 output = self.forward(*inputs, **kwargs)
+                outs = [
+                    o for o in _flatten_outputs(output) if o.requires_grad
+                ]
+                if outs:
+                    torch.autograd.backward([o.sum() for o in outs])
             finally:
                 cfg["shadowing"] = False
-            outs = [o for o in _flatten_outputs(output) if o.requires_grad]
-            if outs:
-                torch.autograd.backward([o.sum() for o in outs])
+                self.require_backward_grad_sync = old_sync
 
     def register_comm_hook(self, state, hook):
         """Gradient-communication hook (stock register_comm_hook).

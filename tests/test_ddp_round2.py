@@ -162,6 +162,41 @@ def test_join_divide_by_active(tmp_path):
     _spawn("_join_div_by_active_body", tmp_path, world=2)
 
 
+def _join_no_sync_body(rank, world):
+    """join() + no_sync() (gradient accumulation with uneven inputs): the
+    S11 agreement flag keeps shadow ranks from posting bucket all-reduces
+    on no-sync iterations — no hang, identical params after."""
+    import msbn
+
+    torch.manual_seed(13)
+    net = msbn.convert_sync_batchnorm(msbn.models.SimpleCNN(width=8))
+    net = msbn.parallel.DistributedDataParallel(net)
+    opt = torch.optim.SGD(net.parameters(), lr=0.05)
+
+    n_batches = 4 if rank == 0 else 2  # both even: accumulation pairs align
+    data = [
+        torch.randn(2, 3, 8, 8,
+                    generator=torch.Generator().manual_seed(rank * 7 + i))
+        for i in range(n_batches)
+    ]
+    with net.join():
+        for i in range(0, len(data), 2):
+            opt.zero_grad(set_to_none=True)
+            with net.no_sync():  # accumulate locally
+                net(data[i]).float().pow(2).mean().backward()
+            net(data[i + 1]).float().pow(2).mean().backward()
+            opt.step()
+
+    flat = torch.cat([p.detach().flatten() for p in net.module.parameters()])
+    flat0 = flat.clone()
+    dist.broadcast(flat0, src=0)
+    assert torch.allclose(flat, flat0, atol=1e-6)
+
+
+def test_join_with_no_sync(tmp_path):
+    _spawn("_join_no_sync_body", tmp_path, world=2)
+
+
 # ------------------------------------------------------- comm hook identity
 def _comm_hook_identity_body(rank, world):
     import msbn
