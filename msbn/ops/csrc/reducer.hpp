@@ -238,7 +238,18 @@ class Reducer : public std::enable_shared_from_this<Reducer> {
     attach_hooks();
   }
 
-  ~Reducer() { *alive_ = false; }
+  ~Reducer() {
+    *alive_ = false;
+    // The last shared_ptr ref can drop on an autograd engine thread (the
+    // queued finalize lambda holds one) — python comm-hook objects must be
+    // released under the GIL.
+    if (py_state_ || py_hook_ || py_bucket_cls_) {
+      py::gil_scoped_acquire gil;
+      py_state_ = py::object();
+      py_hook_ = py::object();
+      py_bucket_cls_ = py::object();
+    }
+  }
 
   void prepare_for_backward(const std::vector<int64_t>& unused_params) {
     std::lock_guard<std::mutex> lock(mutex_);
