@@ -130,3 +130,34 @@ def _fused_world2_body(rank):
 
 def test_fused_syncbn_world2_gpu(tmp_path):
     _spawn("_fused_world2_body", tmp_path)
+
+
+def _join_ctx_gpu_body(rank):
+    """Stock-style `with ddp.join():` on GPU tensors (2 gloo ranks on one
+    device): shadow steps run the 0-batch HIP path (zero-count stats masked
+    in-kernel), final model broadcast from the most-iterated rank."""
+    import msbn
+
+    dev = torch.device("cuda:0")
+    torch.manual_seed(17)
+    net = msbn.convert_sync_batchnorm(msbn.models.SimpleCNN(width=8)).to(dev)
+    net = msbn.parallel.DistributedDataParallel(net)
+    opt = torch.optim.SGD(net.parameters(), lr=0.05)
+    n_batches = 3 if rank == 0 else 1
+    with net.join():
+        for i in range(n_batches):
+            x = torch.randn(
+                2, 3, 8, 8,
+                generator=torch.Generator().manual_seed(rank * 31 + i),
+            ).to(dev)
+            opt.zero_grad(set_to_none=True)
+            net(x).float().pow(2).mean().backward()
+            opt.step()
+    flat = torch.cat([p.detach().flatten() for p in net.module.parameters()])
+    flat0 = flat.clone()
+    dist.broadcast(flat0, src=0)
+    assert torch.allclose(flat, flat0, atol=1e-5)
+
+
+def test_join_context_world2_gpu(tmp_path):
+    _spawn("_join_ctx_gpu_body", tmp_path)
