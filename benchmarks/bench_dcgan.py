@@ -29,6 +29,9 @@ def main():
     p.add_argument("--stock", action="store_true",
                    help="stock torch SyncBatchNorm+DDP comparison line "
                         "(identical architecture and init)")
+    p.add_argument("--benchmark", action="store_true",
+                   help="MIOpen conv autotune (uses/extends the in-tree "
+                        "miopen_db find-db)")
     p.add_argument("--graph", action="store_true",
                    help="capture the G+D step in one hipGraph (msbn only; "
                         "the GAN regime is launch-bound)")
@@ -36,6 +39,13 @@ def main():
                    default=int(os.environ.get("LOCAL_RANK", 0)),
                    dest="local_rank")
     args = p.parse_args()
+
+    _db = os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "miopen_db")
+    if os.path.isdir(_db) and os.listdir(_db):
+        os.environ.setdefault("MIOPEN_USER_DB_PATH", _db)
+    if args.benchmark:
+        torch.backends.cudnn.benchmark = True
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
