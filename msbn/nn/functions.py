@@ -295,6 +295,10 @@ class SyncBatchNormFunction(torch.autograd.Function):
             combined = torch.zeros(
                 2 * C, dtype=torch.float32, device=grad_output.device
             )
+            from msbn.utils import debug as _dbg
+            if _dbg.enabled():  # peers verify; an absent call would hang them
+                _dbg.verify_collective("syncbn.bwd.all_reduce", combined,
+                                       process_group)
             dist.all_reduce(combined, dist.ReduceOp.SUM, group=process_group)
             grad_input = torch.empty_like(grad_output)
 

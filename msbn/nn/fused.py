@@ -148,6 +148,10 @@ class SyncBatchNormActFunction(torch.autograd.Function):
             if world_size > 1 and (need_input_g or need_res_g):
                 combined = torch.zeros(2 * C, dtype=torch.float32,
                                        device=grad_output.device)
+                from msbn.utils import debug as _dbg
+                if _dbg.enabled():  # peers verify; skipping would hang them
+                    _dbg.verify_collective("syncbn.bwd.all_reduce", combined,
+                                           process_group)
                 dist.all_reduce(combined, dist.ReduceOp.SUM,
                                 group=process_group)
             zg = torch.empty_like(grad_output)

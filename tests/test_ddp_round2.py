@@ -530,6 +530,27 @@ def test_debug_collectives_mismatch(tmp_path):
     _spawn("_debug_collectives_mismatch_body", tmp_path, world=2)
 
 
+def _debug_collectives_empty_rank_body(rank, world):
+    """Verification also fires on empty-input ranks (their zero-contribution
+    collectives must verify too, or the peers' side-group gather hangs)."""
+    import msbn
+
+    os.environ["MSBN_DEBUG_COLLECTIVES"] = "1"
+    try:
+        bn = msbn.nn.SyncBatchNorm(4)
+        bn.train()
+        n = 3 if rank == 0 else 0
+        x = torch.randn(n, 4, 5, 5, requires_grad=True)
+        y = bn(x)
+        (y.sum() * 1.0).backward()
+    finally:
+        os.environ.pop("MSBN_DEBUG_COLLECTIVES", None)
+
+
+def test_debug_collectives_empty_rank(tmp_path):
+    _spawn("_debug_collectives_empty_rank_body", tmp_path, world=2)
+
+
 # ----------------------------------------------------------- backward stats
 def _backward_stats_body(rank, world):
     """Reducer records per-param grad-ready timestamps (stock
