@@ -91,6 +91,7 @@ class SyncBatchNormActFunction(torch.autograd.Function):
         ctx.relu = relu
         ctx.process_group = process_group
         ctx.world_size = world_size
+        ctx.use_sync = use_sync
         if input.numel() == 0:
             return torch.empty_like(input)
         return ops.batch_norm_elemt_act(
@@ -143,9 +144,10 @@ class SyncBatchNormActFunction(torch.autograd.Function):
             relu and residual is not None and need_res_g and input.is_cuda
             and input.numel() > 0
         )
+        use_sync = getattr(ctx, "use_sync", world_size > 1)
         if input.numel() == 0:
             # empty-input rank (join): contribute zeros, keep peers unblocked
-            if world_size > 1 and (need_input_g or need_res_g):
+            if use_sync and (need_input_g or need_res_g):
                 combined = torch.zeros(2 * C, dtype=torch.float32,
                                        device=grad_output.device)
                 from msbn.utils import debug as _dbg
@@ -171,7 +173,7 @@ class SyncBatchNormActFunction(torch.autograd.Function):
         )
         grad_input = grad_res = None
         if need_input_g or need_res_g:
-            if world_size > 1:
+            if use_sync:
                 combined, copied = _combined_view(sum_dy, sum_dy_xmu, C)
                 from msbn.utils import debug as _dbg
                 if _dbg.enabled():
