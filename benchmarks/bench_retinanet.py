@@ -72,15 +72,12 @@ def main():
                                 rank=rank)
 
     torch.manual_seed(11)
-    import contextlib
-
     use_graph = args.graph and use_cuda and not args.stock
     # NOTE: warmup AND capture run on `side`; grad accumulators + optimizer
     # state are created during warmup, so they live on the capture stream
     # (the DDP-under-graph recipe, tools/nccl_world1_check.py).  At world>1
     # wrap DDP construction in torch.cuda.stream(side) as bench.py does.
     side = torch.cuda.Stream() if use_graph else None
-    del contextlib
     if args.stock:
         from msbn.models import convert_to_torch_batchnorm
 
