@@ -49,6 +49,9 @@ def one_case(rng):
     with_res = rng.random() < 0.5
     desc = f"shape={shape} dtype={dtype} cl={cl} relu={relu} res={with_res}"
     one_case.last = desc
+    if os.environ.get("MSBN_FUZZ_VERBOSE", "0") == "1":
+        print("CASE", desc, flush=True)
+        torch.cuda.synchronize()  # attribute any fault to the printed case
 
     fmt = (torch.channels_last if dims == 4 else torch.channels_last_3d) \
         if cl else torch.contiguous_format
