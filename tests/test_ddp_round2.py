@@ -197,6 +197,62 @@ def test_join_with_no_sync(tmp_path):
     _spawn("_join_no_sync_body", tmp_path, world=2)
 
 
+def _join_interactions_body(rank, world):
+    """join() composed with find_unused_parameters, wire compression, and a
+    custom Python comm hook — every combination must stay matched across
+    uneven ranks."""
+    import msbn
+    from torch.distributed.algorithms.ddp_comm_hooks import default_hooks as dh
+
+    class Branchy(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.a = torch.nn.Linear(6, 6)
+            self.b = torch.nn.Linear(6, 6)  # unused every iteration
+
+        def forward(self, x):
+            return self.a(x)
+
+    def train(hook=None, model_cls=None, **ddp_kw):
+        torch.manual_seed(21)
+        model = (model_cls or torch.nn.Linear)(6, 6) \
+            if model_cls is not Branchy else Branchy()
+        net = msbn.parallel.DistributedDataParallel(model, **ddp_kw)
+        if hook is not None:
+            net.register_comm_hook(None, hook)
+        opt = torch.optim.SGD(net.parameters(), lr=0.05)
+        n = 3 if rank == 0 else 1
+        with net.join():
+            for i in range(n):
+                x = torch.randn(
+                    2, 6, generator=torch.Generator().manual_seed(rank + i))
+                opt.zero_grad(set_to_none=True)
+                net(x).pow(2).mean().backward()
+                opt.step()
+        flat = torch.cat(
+            [p.detach().flatten() for p in net.module.parameters()])
+        ref = flat.clone()
+        dist.broadcast(ref, src=0)
+        assert torch.allclose(flat, ref, atol=1e-6)
+
+    # unused-param model needs find_unused or static_graph
+    train(model_cls=Branchy, find_unused_parameters=True)
+    train(model_cls=Branchy, static_graph=True)
+    train(hook=dh.bf16_compress_hook)
+
+    def my_hook(state, bucket):
+        t = bucket.buffer()
+        t.div_(world)
+        work = dist.all_reduce(t, async_op=True)
+        return work.get_future().then(lambda f: f.value()[0])
+
+    train(hook=my_hook)
+
+
+def test_join_interactions(tmp_path):
+    _spawn("_join_interactions_body", tmp_path, world=2)
+
+
 # ------------------------------------------------------- comm hook identity
 def _comm_hook_identity_body(rank, world):
     import msbn
